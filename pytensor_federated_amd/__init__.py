@@ -19,6 +19,15 @@ designed MI355X-first:
 Like the reference's ``__init__.py:1-12``, graph-layer imports are optional
 so the transport stack works standalone.
 """
+import os as _os
+
+# grpc's default epoll1 poller aborts forked children; these must be set
+# before the first `import grpc` anywhere in the process.  Needed for parity
+# with the reference's fork-based multiprocessing support (PyMC chain
+# workers re-connect after fork via the `thread_pid_id` keying).
+_os.environ.setdefault("GRPC_ENABLE_FORK_SUPPORT", "true")
+_os.environ.setdefault("GRPC_POLL_STRATEGY", "epoll1")
+
 from . import npproto, rpc  # noqa: F401
 from .common import (  # noqa: F401
     LogpGradServiceClient,

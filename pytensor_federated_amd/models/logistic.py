@@ -1,0 +1,115 @@
+"""Logistic-GLM logp+grad model (BASELINE.json config 4 family).
+
+The GLM generalization of the reference's ComputeFunc contract
+(signatures.py:27-33): parameters ``beta[K]``, private shard ``X[N,K]``,
+``y[N] in {0,1}``:
+
+    z    = X @ beta
+    logp = sum( y*z - softplus(z) )          (numerically stable BCE)
+    grad = X^T @ (y - sigmoid(z))
+
+Compute paths:
+* eager torch (CPU fallback / golden reference): two rocBLAS matmuls +
+  elementwise;
+* MI355X: ONE fused CDNA4 HIP kernel reading X exactly once -- per row-tile
+  it computes z, sigmoid, the logp term and the rank-1 grad update while the
+  tile is still in registers.  At N=1e8 x K=1024 bf16 the shard is 25.6 GB
+  per GPU; the op is HBM-bound, so single-pass is ~2x over the eager
+  two-matmul shape.
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Tuple
+
+import numpy as np
+import torch
+
+from .base import LogpGradModel
+
+__all__ = ["LogisticGLMModel", "generate_logistic_dataset"]
+
+
+def generate_logistic_dataset(
+    n_rows: int,
+    n_features: int,
+    *,
+    seed: int = 0,
+    beta_scale: float = 0.5,
+) -> Tuple[np.ndarray, np.ndarray, np.ndarray]:
+    """Synthetic (X, y, beta_true) with ~balanced classes."""
+    rng = np.random.RandomState(seed)
+    X = rng.standard_normal((n_rows, n_features)) / np.sqrt(n_features)
+    beta = rng.standard_normal(n_features) * beta_scale
+    p = 1.0 / (1.0 + np.exp(-(X @ beta)))
+    y = (rng.uniform(size=n_rows) < p).astype(np.float64)
+    return X, y, beta
+
+
+class LogisticGLMModel(LogpGradModel):
+    """Federated worker model: logistic regression on a private shard."""
+
+    param_names = ("beta",)
+
+    def __init__(
+        self,
+        X,
+        y,
+        *,
+        device=None,
+        dtype: torch.dtype = None,
+        use_kernels: Optional[bool] = None,
+        delay: Optional[float] = None,
+    ) -> None:
+        super().__init__(delay=delay)
+        X = torch.as_tensor(np.asarray(X)) if not isinstance(X, torch.Tensor) else X
+        y = torch.as_tensor(np.asarray(y)) if not isinstance(y, torch.Tensor) else y
+        if dtype is None:
+            dtype = X.dtype if X.is_floating_point() else torch.float64
+        if device is None:
+            device = X.device
+        self._X = X.to(device=device, dtype=dtype).contiguous()
+        self._y = y.to(device=device, dtype=dtype).contiguous()
+        self._use_kernels = use_kernels
+        if self._X.dim() != 2 or self._y.dim() != 1 or self._X.shape[0] != self._y.shape[0]:
+            raise ValueError("X must be [N,K] and y [N].")
+        self._n, self._k = self._X.shape
+
+    @property
+    def device(self):
+        return self._X.device
+
+    @property
+    def n_rows(self) -> int:
+        return int(self._n)
+
+    @property
+    def n_features(self) -> int:
+        return int(self._k)
+
+    def _kernel_path(self) -> bool:
+        if self._use_kernels is None:
+            return self._X.is_cuda
+        return self._use_kernels
+
+    def logp_grad(self, beta) -> Tuple[torch.Tensor, List[torch.Tensor]]:
+        beta = torch.as_tensor(beta)
+        if beta.shape != (self._k,):
+            raise ValueError(f"beta must have shape ({self._k},), got {tuple(beta.shape)}.")
+        if self._kernel_path():
+            from ..ops import logistic_glm_logp_grad
+
+            logp, grad = logistic_glm_logp_grad(self._X, self._y, beta)
+            return logp, [grad]
+        return self._logp_grad_eager(beta)
+
+    def _logp_grad_eager(self, beta: torch.Tensor) -> Tuple[torch.Tensor, List[torch.Tensor]]:
+        X, y = self._X, self._y
+        acc_dtype = torch.float64 if X.dtype == torch.float64 else torch.float32
+        beta = beta.to(device=X.device, dtype=acc_dtype)
+        Xf = X.to(acc_dtype)
+        yf = y.to(acc_dtype)
+        z = Xf @ beta
+        logp = torch.sum(yf * z - torch.nn.functional.softplus(z), dtype=torch.float64)
+        resid = yf - torch.sigmoid(z)
+        grad = Xf.t() @ resid
+        return logp, [grad.to(torch.float64)]

@@ -1,0 +1,187 @@
+"""ODE logp model with adjoint gradients (BASELINE.json config 5).
+
+``[theta] -> trajectories -> Gaussian log-likelihood`` with the gradient
+computed by the discrete adjoint method: one RK4 forward sweep storing only
+the step states, then a backward sweep propagating the adjoint state via
+per-step vector-Jacobian products (torch.autograd re-derives each step's
+local Jacobian action; memory is O(n_steps * state), never O(n_steps *
+graph)).  The discrete adjoint is exact for the discretized system, so
+federated workers (each owning its private experiments batch) return
+gradients that sum exactly across shards -- same identity the linear/GLM
+models exploit.
+
+Batching: each worker integrates a whole batch ``u[B, D]`` of experiments
+in one vectorized sweep -- on an MI355X the RK4 right-hand side evaluates
+as batched torch tensor ops on the GPU.
+"""
+from __future__ import annotations
+
+import math
+from typing import Callable, List, Optional, Tuple
+
+import numpy as np
+import torch
+
+from .base import LogpGradModel
+
+__all__ = ["ODEModel", "lotka_volterra_rhs", "generate_ode_dataset"]
+
+
+def lotka_volterra_rhs(t, u: torch.Tensor, theta: torch.Tensor) -> torch.Tensor:
+    """Classic 2-species Lotka-Volterra; u=[B,2], theta=[alpha,beta,gamma,delta]."""
+    prey, pred = u[..., 0], u[..., 1]
+    alpha, beta, gamma, delta = theta[0], theta[1], theta[2], theta[3]
+    dprey = alpha * prey - beta * prey * pred
+    dpred = delta * prey * pred - gamma * pred
+    return torch.stack([dprey, dpred], dim=-1)
+
+
+def _rk4_step(f, t, u, h, theta):
+    k1 = f(t, u, theta)
+    k2 = f(t + 0.5 * h, u + 0.5 * h * k1, theta)
+    k3 = f(t + 0.5 * h, u + 0.5 * h * k2, theta)
+    k4 = f(t + h, u + h * k3, theta)
+    return u + (h / 6.0) * (k1 + 2 * k2 + 2 * k3 + k4)
+
+
+class ODEModel(LogpGradModel):
+    """Gaussian LL of observed ODE trajectories, adjoint gradient w.r.t. theta."""
+
+    param_names = ("theta",)
+
+    def __init__(
+        self,
+        f: Callable,
+        u0,
+        t0: float,
+        t1: float,
+        n_steps: int,
+        obs_indices,
+        y_obs,
+        sigma: float,
+        *,
+        obs_components: Optional[list] = None,
+        device=None,
+        dtype: torch.dtype = torch.float64,
+        delay: Optional[float] = None,
+    ) -> None:
+        """
+        Parameters
+        ----------
+        f : callable(t, u, theta) -> du/dt
+            Vectorized RHS over the experiments batch ``u[B, D]``.
+        u0 : tensor [B, D]
+            Initial states of this worker's private experiments.
+        t0, t1, n_steps
+            Fixed-step RK4 grid (h = (t1-t0)/n_steps).
+        obs_indices : int array [n_obs]
+            Step indices (0..n_steps) at which observations were taken.
+        y_obs : tensor [n_obs, B, n_comp]
+            Observed values (private data shard).
+        sigma : float
+            Observation noise scale.
+        obs_components : list of int, optional
+            Which state components are observed (default: all).
+        """
+        super().__init__(delay=delay)
+        self.f = f
+        u0 = torch.as_tensor(np.asarray(u0)) if not isinstance(u0, torch.Tensor) else u0
+        y = torch.as_tensor(np.asarray(y_obs)) if not isinstance(y_obs, torch.Tensor) else y_obs
+        if device is None:
+            device = u0.device
+        self._u0 = u0.to(device=device, dtype=dtype)
+        self._y = y.to(device=device, dtype=dtype)
+        self._t0 = float(t0)
+        self._h = (float(t1) - float(t0)) / int(n_steps)
+        self._n_steps = int(n_steps)
+        self._obs_idx = [int(i) for i in obs_indices]
+        if sorted(self._obs_idx) != self._obs_idx:
+            raise ValueError("obs_indices must be sorted ascending.")
+        self._obs_components = obs_components
+        self._sigma = float(sigma)
+        self._dtype = dtype
+
+    @property
+    def device(self):
+        return self._u0.device
+
+    def _project(self, u: torch.Tensor) -> torch.Tensor:
+        if self._obs_components is None:
+            return u
+        return u[..., self._obs_components]
+
+    def _forward_states(self, theta: torch.Tensor) -> List[torch.Tensor]:
+        states = [self._u0]
+        u = self._u0
+        with torch.no_grad():
+            for k in range(self._n_steps):
+                t = self._t0 + k * self._h
+                u = _rk4_step(self.f, t, u, self._h, theta)
+                states.append(u)
+        return states
+
+    def logp_grad(self, theta) -> Tuple[torch.Tensor, List[torch.Tensor]]:
+        theta = torch.as_tensor(theta).to(device=self.device, dtype=self._dtype)
+        states = self._forward_states(theta)
+
+        sig2 = self._sigma * self._sigma
+        n_obs_values = 0
+        logp = torch.zeros((), dtype=torch.float64, device=self.device)
+        # dL/du at each observation (fed into the adjoint when the backward
+        # sweep passes that step index)
+        obs_grad = {}
+        for j, idx in enumerate(self._obs_idx):
+            pred = self._project(states[idx])
+            resid = self._y[j] - pred
+            n_obs_values += resid.numel()
+            logp = logp - torch.sum(resid * resid, dtype=torch.float64) / (2.0 * sig2)
+            g = torch.zeros_like(states[idx])
+            if self._obs_components is None:
+                g += resid / sig2
+            else:
+                g[..., self._obs_components] = resid / sig2
+            obs_grad[idx] = g
+        logp = logp - 0.5 * n_obs_values * math.log(2.0 * math.pi * sig2)
+
+        # ---- discrete adjoint backward sweep -------------------------
+        lam = torch.zeros_like(self._u0)
+        g_theta = torch.zeros_like(theta)
+        if self._n_steps in obs_grad:
+            lam = lam + obs_grad[self._n_steps]
+        for k in range(self._n_steps - 1, -1, -1):
+            t = self._t0 + k * self._h
+            u_k = states[k].detach().requires_grad_(True)
+            theta_k = theta.detach().requires_grad_(True)
+            with torch.enable_grad():
+                u_next = _rk4_step(self.f, t, u_k, self._h, theta_k)
+                gu, gth = torch.autograd.grad(u_next, (u_k, theta_k), grad_outputs=lam)
+            g_theta = g_theta + gth
+            lam = gu
+            if k in obs_grad:
+                lam = lam + obs_grad[k]
+        return logp, [g_theta.to(torch.float64)]
+
+
+def generate_ode_dataset(
+    n_experiments: int = 8,
+    n_obs: int = 20,
+    n_steps: int = 100,
+    t1: float = 10.0,
+    sigma: float = 0.1,
+    theta=(0.8, 0.3, 0.6, 0.2),
+    seed: int = 0,
+):
+    """Synthetic Lotka-Volterra observations for one worker's shard."""
+    rng = np.random.RandomState(seed)
+    u0 = 1.0 + rng.uniform(0.0, 1.0, size=(n_experiments, 2))
+    theta_t = torch.as_tensor(np.asarray(theta, dtype=np.float64))
+    u = torch.as_tensor(u0)
+    h = t1 / n_steps
+    obs_idx = np.linspace(1, n_steps, n_obs).astype(int).tolist()
+    states = [u]
+    for k in range(n_steps):
+        u = _rk4_step(lotka_volterra_rhs, k * h, u, h, theta_t)
+        states.append(u)
+    y = np.stack([states[i].numpy() for i in obs_idx])
+    y += rng.normal(scale=sigma, size=y.shape)
+    return u0, obs_idx, y

@@ -1,0 +1,157 @@
+"""Python interface to the CDNA4 HIP kernels.
+
+Loads the in-tree ``libfedops_gfx950.so`` via ctypes (no torch C++ ABI
+coupling; launches go onto torch's current HIP stream, so the ops compose
+with streams/graphs and RCCL).  On a GPU box a missing extension is a HARD
+error -- there is deliberately no silent eager fallback here (the eager
+path is an explicit model-level opt-out, ``use_kernels=False``).
+"""
+from __future__ import annotations
+
+import ctypes
+from typing import Optional, Tuple
+
+import torch
+
+from .build import LIB_PATH
+
+__all__ = [
+    "kernels_available",
+    "require_kernels",
+    "gaussian_linear_logp_grad",
+    "logistic_glm_logp_grad",
+]
+
+_FED_F32, _FED_F64, _FED_BF16 = 0, 1, 2
+_DTYPE_CODE = {
+    torch.float32: _FED_F32,
+    torch.float64: _FED_F64,
+    torch.bfloat16: _FED_BF16,
+}
+
+_lib: Optional[ctypes.CDLL] = None
+_load_error: Optional[str] = None
+
+
+def _try_load() -> Optional[ctypes.CDLL]:
+    global _lib, _load_error
+    if _lib is not None:
+        return _lib
+    path = LIB_PATH
+    if not path.exists():
+        # Cross-compile on the fly when hipcc is present (seconds on CPU box).
+        try:
+            from .build import build
+
+            build()
+        except Exception as ex:  # no hipcc / compile error
+            _load_error = f"extension not built and build failed: {ex}"
+            return None
+    try:
+        lib = ctypes.CDLL(str(path))
+    except OSError as ex:
+        _load_error = f"failed to dlopen {path}: {ex}"
+        return None
+    lib.fed_gaussian_linear.restype = ctypes.c_int
+    lib.fed_gaussian_linear.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_longlong,
+        ctypes.c_double, ctypes.c_double, ctypes.c_double,
+        ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
+    ]
+    lib.fed_logistic_glm.restype = ctypes.c_int
+    lib.fed_logistic_glm.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_longlong, ctypes.c_int,
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_longlong,
+        ctypes.c_int, ctypes.c_void_p,
+    ]
+    lib.fed_last_hip_error.restype = ctypes.c_char_p
+    _lib = lib
+    return _lib
+
+
+def kernels_available() -> bool:
+    return _try_load() is not None
+
+
+def require_kernels() -> ctypes.CDLL:
+    lib = _try_load()
+    if lib is None:
+        raise RuntimeError(
+            "pytensor_federated_amd HIP extension is not available on this GPU box "
+            f"({_load_error}). Build it with `python -m pytensor_federated_amd.ops.build` "
+            "or run __graft_entry__.build(). Refusing to fall back to eager torch "
+            "on a GPU (pass use_kernels=False to the model to opt out explicitly)."
+        )
+    return lib
+
+
+def _stream_ptr() -> int:
+    return torch.cuda.current_stream().cuda_stream
+
+
+def _check(rc: int, what: str) -> None:
+    if rc != 0:
+        lib = _try_load()
+        detail = lib.fed_last_hip_error().decode() if lib is not None else "?"
+        raise RuntimeError(f"{what} failed with code {rc} ({detail})")
+
+
+def gaussian_linear_logp_grad(
+    x: torch.Tensor,
+    y: torch.Tensor,
+    a: float,
+    b: float,
+    sigma: float,
+    out: Optional[torch.Tensor] = None,
+) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Fused logp + dlogp/da + dlogp/db on device; returns fp64 0-d views.
+
+    ``out`` (fp64[3] on the same device) may be supplied to keep the result
+    buffer stable across calls -- it is the buffer an RCCL all-reduce sums
+    in the federated path.  Fully async on the current stream.
+    """
+    lib = require_kernels()
+    if x.dtype not in _DTYPE_CODE:
+        raise TypeError(f"unsupported dtype {x.dtype}")
+    assert x.is_cuda and y.is_cuda and x.is_contiguous() and y.is_contiguous()
+    if out is None:
+        out = torch.empty(3, dtype=torch.float64, device=x.device)
+    rc = lib.fed_gaussian_linear(
+        x.data_ptr(), y.data_ptr(), x.numel(),
+        float(a), float(b), float(sigma),
+        out.data_ptr(), _DTYPE_CODE[x.dtype], _stream_ptr(),
+    )
+    _check(rc, "fed_gaussian_linear")
+    return out[0], out[1], out[2]
+
+
+_ws_cache = {}
+
+
+def logistic_glm_logp_grad(
+    X: torch.Tensor,
+    y: torch.Tensor,
+    beta: torch.Tensor,
+    out: Optional[torch.Tensor] = None,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Fused logistic-GLM logp+grad; X read once. Returns (logp, grad[K]) fp64 views."""
+    lib = require_kernels()
+    n, K = X.shape
+    if X.dtype not in (torch.bfloat16, torch.float32):
+        raise TypeError(f"unsupported dtype {X.dtype}")
+    assert X.is_cuda and X.is_contiguous()
+    beta_f32 = beta.detach().to(device=X.device, dtype=torch.float32).contiguous()
+    if out is None:
+        out = torch.empty(1 + K, dtype=torch.float64, device=X.device)
+    key = (X.device.index, K)
+    ws = _ws_cache.get(key)
+    if ws is None:
+        ws = torch.empty(1024 * K, dtype=torch.float32, device=X.device)
+        _ws_cache[key] = ws
+    rc = lib.fed_logistic_glm(
+        X.data_ptr(), y.data_ptr(), n, K,
+        beta_f32.data_ptr(), out.data_ptr(), ws.data_ptr(),
+        ws.numel() * 4, _DTYPE_CODE[X.dtype], _stream_ptr(),
+    )
+    _check(rc, "fed_logistic_glm")
+    return out[0], out[1:]

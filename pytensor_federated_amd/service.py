@@ -293,6 +293,23 @@ class ClientPrivates:
 _privates: Dict[str, ClientPrivates] = {}
 
 
+def _after_fork_in_child() -> None:
+    """Give forked children a fresh event loop.
+
+    A forked child inherits the parent's loop object (shared self-pipe fds,
+    possibly held locks) -- unusable.  Connections re-establish lazily via
+    the pid-keyed ``_privates`` cache, which is why clients survive
+    ``multiprocessing`` with both fork and spawn (PyMC chain workers)."""
+    try:
+        asyncio.set_event_loop(asyncio.new_event_loop())
+    except Exception:
+        pass
+
+
+if hasattr(os, "register_at_fork"):
+    os.register_at_fork(after_in_child=_after_fork_in_child)
+
+
 def thread_pid_id(obj: object) -> str:
     """Identifier unique to (object, process, thread) (reference service.py:273-275)."""
     return f"{id(obj)}-{os.getpid()}-{threading.get_ident()}"
