@@ -1,0 +1,188 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: federated logp+grad calls/sec (BASELINE.json metric).
+
+Measures the linear-regression demo config on 1..8 MI355X GPUs:
+each rank (GPU) owns a private synthetic shard of 1e7 bf16 rows; one
+"step" = one full federated logp+grad evaluation -- broadcast theta,
+fused HIP kernel per shard, RCCL all-reduce of [logp, d/da, d/db], result
+read back to the driver (what a PyMC client would receive).
+
+Launch (the driver does this):
+  python bench.py --gpus 1 --steps K --warmup W
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Rank 0 prints ONE JSON line with the whole-node call rate
+(value = evals/sec * n_gpus: each federated evaluation is one worker-call
+per GPU shard; weak scaling -- per-GPU rows fixed as N grows).
+"""
+import argparse
+import json
+import math
+import os
+import time
+
+import numpy as np
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=200)
+    p.add_argument("--warmup", type=int, default=20)
+    p.add_argument("--model", choices=["linear", "logistic"], default="linear")
+    p.add_argument("--rows", type=int, default=None, help="rows per GPU shard")
+    p.add_argument("--features", type=int, default=1024, help="logistic GLM features")
+    p.add_argument("--dtype", choices=["bf16", "f32", "f64"], default="bf16")
+    p.add_argument("--eager", action="store_true", help="disable HIP kernels (debug)")
+    p.add_argument(
+        "--no-readback",
+        action="store_true",
+        help="skip per-step host readback (measures enqueue throughput only; "
+        "NOT the headline metric)",
+    )
+    return p.parse_known_args()[0]
+
+
+def main():
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    distributed = world > 1
+
+    have_gpu = torch.cuda.is_available()
+    if have_gpu:
+        local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+    else:
+        device = torch.device("cpu")
+
+    dtype = {"bf16": torch.bfloat16, "f32": torch.float32, "f64": torch.float64}[args.dtype]
+    if not have_gpu and dtype == torch.bfloat16:
+        dtype = torch.float32  # CPU plumbing check only
+    rows = args.rows
+    if rows is None:
+        rows = 10_000_000 if have_gpu else 100_000
+        if args.model == "logistic":
+            rows = 12_500_000 if have_gpu else 10_000
+
+    if distributed:
+        import torch.distributed as dist
+
+        dist.init_process_group(backend="nccl" if have_gpu else "gloo")
+
+    from pytensor_federated_amd.models import GaussianLinearModel, LogisticGLMModel
+    from pytensor_federated_amd.parallel import FederatedShardEngine
+
+    use_kernels = None if (have_gpu and not args.eager) else False
+
+    # Private per-rank shard: synthetic, seeded by rank (federated = every
+    # worker owns different data).  Generate on device to keep startup fast.
+    gen = torch.Generator(device="cpu").manual_seed(1234 + rank)
+    if args.model == "linear":
+        x = torch.rand(rows, generator=gen) * 10.0
+        noise = torch.randn(rows, generator=gen) * 0.4
+        y = 1.5 + 0.5 * x + noise
+        model = GaussianLinearModel(
+            x, y, sigma=0.4, device=device, dtype=dtype, use_kernels=use_kernels
+        )
+        theta0 = np.array([1.5, 0.5])
+        config_model = "gaussian_linear_regression"
+        shape_cfg = {"rows_per_gpu": rows}
+    else:
+        k = args.features
+        X = torch.randn(rows, k, generator=gen) / math.sqrt(k)
+        beta_true = torch.randn(k, generator=gen) * 0.5
+        p = torch.sigmoid(X @ beta_true)
+        y = (torch.rand(rows, generator=gen) < p).to(torch.float32)
+        model = LogisticGLMModel(X, y, device=device, dtype=dtype, use_kernels=use_kernels)
+        theta0 = beta_true.numpy()
+        config_model = "logistic_glm"
+        shape_cfg = {"rows_per_gpu": rows, "features": k}
+
+    engine = FederatedShardEngine(model, use_distributed=distributed)
+
+    theta_base = torch.as_tensor(theta0, dtype=torch.float64, device=device)
+    readback = not args.no_readback
+
+    def one_step(t: int):
+        # every rank derives the same perturbed theta (the broadcast of theta
+        # from the driver is folded into the all-reduce round trip below)
+        scale = 1.0 + 0.001 * math.sin(t)
+        if args.model == "linear":
+            logp, grads = engine.logp_grad(
+                float(theta_base[0]) * scale, float(theta_base[1]) * scale
+            )
+        else:
+            logp, grads = engine.logp_grad(theta_base * scale)
+        if readback:
+            # deliver the federated [logp, grads] to the driver like a real
+            # client call would (forces the per-call sync)
+            return engine._buf.cpu()
+        return None
+
+    def barrier_sync():
+        if distributed:
+            import torch.distributed as dist
+
+            dist.barrier()
+        if have_gpu:
+            torch.cuda.synchronize()
+
+    for t in range(args.warmup):
+        one_step(t)
+    barrier_sync()
+    t0 = time.perf_counter()
+    for t in range(args.steps):
+        one_step(t)
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    if distributed:
+        import torch.distributed as dist
+
+        e = torch.tensor([elapsed], dtype=torch.float64, device=device if have_gpu else "cpu")
+        dist.all_reduce(e, op=dist.ReduceOp.MAX)
+        elapsed = float(e[0])
+
+    n_gpus = world if distributed else 1
+    evals_per_s = args.steps / elapsed
+    value = evals_per_s * n_gpus  # whole-node worker-call rate
+    if rank == 0:
+        result = {
+            "metric": "logp+grad calls/sec (whole node)",
+            "value": value,
+            "unit": "calls/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": args.dtype if have_gpu else "f32",
+            "data": "synthetic",
+            "config": {
+                "model": config_model,
+                "parallelism": f"federated shard-dp x{n_gpus} (RCCL all-reduce)",
+                "per_step": "theta -> fused logp+grad kernel per shard -> "
+                "all_reduce([logp,*grads]) -> host readback",
+                "readback": readback,
+                "device": str(device),
+                "kernels": bool(use_kernels is None),
+                **shape_cfg,
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    if distributed:
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
