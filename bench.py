@@ -104,23 +104,29 @@ def main():
 
     engine = FederatedShardEngine(model, use_distributed=distributed)
 
-    theta_base = torch.as_tensor(theta0, dtype=torch.float64, device=device)
     readback = not args.no_readback
+    a0, b0 = (float(theta0[0]), float(theta0[1])) if args.model == "linear" else (0.0, 0.0)
+    theta_dev = torch.as_tensor(theta0, dtype=torch.float64, device=device)
+    host_buf = None
 
     def one_step(t: int):
+        nonlocal host_buf
         # every rank derives the same perturbed theta (the broadcast of theta
         # from the driver is folded into the all-reduce round trip below)
         scale = 1.0 + 0.001 * math.sin(t)
         if args.model == "linear":
-            logp, grads = engine.logp_grad(
-                float(theta_base[0]) * scale, float(theta_base[1]) * scale
-            )
+            buf = engine.logp_grad_fused(a0 * scale, b0 * scale)
         else:
-            logp, grads = engine.logp_grad(theta_base * scale)
+            buf = engine.logp_grad_fused(theta_dev * scale)
         if readback:
             # deliver the federated [logp, grads] to the driver like a real
             # client call would (forces the per-call sync)
-            return engine._buf.cpu()
+            if host_buf is None:
+                host_buf = torch.empty_like(buf, device="cpu", pin_memory=have_gpu)
+            host_buf.copy_(buf)
+            if have_gpu:
+                torch.cuda.synchronize()
+            return host_buf
         return None
 
     def barrier_sync():

@@ -110,15 +110,30 @@ class GaussianLinearModel(LogpGradModel):
             return on_gpu
         return self._use_kernels
 
-    def logp_grad(self, intercept, slope) -> Tuple[torch.Tensor, List[torch.Tensor]]:
+    #: layout of the fused fp64 output buffer: [logp, dlogp/da, dlogp/db]
+    fused_size = 3
+
+    def logp_grad(
+        self, intercept, slope, out: Optional[torch.Tensor] = None
+    ) -> Tuple[torch.Tensor, List[torch.Tensor]]:
+        """``out`` (fp64[3], same device) receives the fused [logp, ga, gb]
+        in place -- the zero-copy seam to the RCCL all-reduce buffer."""
         a = float(intercept)
         b = float(slope)
         if self._kernel_path():
             from ..ops import gaussian_linear_logp_grad
 
-            logp, ga, gb = gaussian_linear_logp_grad(self._x, self._y, a, b, self._sigma)
+            logp, ga, gb = gaussian_linear_logp_grad(
+                self._x, self._y, a, b, self._sigma, out=out
+            )
             return logp, [ga, gb]
-        return self._logp_grad_eager(a, b)
+        logp, grads = self._logp_grad_eager(a, b)
+        if out is not None:
+            out[0] = logp
+            out[1] = grads[0]
+            out[2] = grads[1]
+            return out[0], [out[1], out[2]]
+        return logp, grads
 
     def _logp_grad_eager(self, a: float, b: float) -> Tuple[torch.Tensor, List[torch.Tensor]]:
         x, y = self._x, self._y

@@ -91,16 +91,28 @@ class LogisticGLMModel(LogpGradModel):
             return self._X.is_cuda
         return self._use_kernels
 
-    def logp_grad(self, beta) -> Tuple[torch.Tensor, List[torch.Tensor]]:
+    @property
+    def fused_size(self) -> int:
+        """Layout of the fused fp64 output buffer: [logp, grad[K]]."""
+        return 1 + int(self._k)
+
+    def logp_grad(
+        self, beta, out: Optional[torch.Tensor] = None
+    ) -> Tuple[torch.Tensor, List[torch.Tensor]]:
         beta = torch.as_tensor(beta)
         if beta.shape != (self._k,):
             raise ValueError(f"beta must have shape ({self._k},), got {tuple(beta.shape)}.")
         if self._kernel_path():
             from ..ops import logistic_glm_logp_grad
 
-            logp, grad = logistic_glm_logp_grad(self._X, self._y, beta)
+            logp, grad = logistic_glm_logp_grad(self._X, self._y, beta, out=out)
             return logp, [grad]
-        return self._logp_grad_eager(beta)
+        logp, grads = self._logp_grad_eager(beta)
+        if out is not None:
+            out[0] = logp
+            out[1:] = grads[0]
+            return out[0], [out[1:]]
+        return logp, grads
 
     def _logp_grad_eager(self, beta: torch.Tensor) -> Tuple[torch.Tensor, List[torch.Tensor]]:
         X, y = self._X, self._y

@@ -139,37 +139,42 @@ __global__ __launch_bounds__(256) void k_gaussian_linear(
     using TR = VecTraits<T>;
     using A = typename TR::acc_t;
     constexpr int VEC = TR::VEC;
-    const A a = (A)a_d;
-    const A b = (A)b_d;
+    // Element math + accumulation in fp64 for every input dtype: the kernel
+    // is HBM-bound (4 B/row bf16), so the extra f64 FMAs are free, and the
+    // cancellation-heavy sum(r) would otherwise carry a systematic fp32
+    // pred-rounding bias (~1e-3 relative at N=1e7 -- measured).
+    const double a = a_d;
+    const double b = b_d;
 
     const long long gid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
     const long long gstride = (long long)gridDim.x * blockDim.x;
     const long long nvec = n / VEC;
 
-    A sr = 0, srx = 0, sr2 = 0;
+    double sr = 0, srx = 0, sr2 = 0;
     A xv[VEC], yv[VEC];
     for (long long i = gid; i < nvec; i += gstride) {
         TR::load(x + i * VEC, xv);
         TR::load(y + i * VEC, yv);
 #pragma unroll
         for (int j = 0; j < VEC; ++j) {
-            const A r = yv[j] - (a + b * xv[j]);
+            const double xd = (double)xv[j];
+            const double r = (double)yv[j] - (a + b * xd);
             sr += r;
-            srx += r * xv[j];
+            srx += r * xd;
             sr2 += r * r;
         }
     }
     // scalar tail
     for (long long i = nvec * VEC + gid; i < n; i += gstride) {
-        const A xi = TR::get(x, i);
-        const A r = TR::get(y, i) - (a + b * xi);
+        const double xi = (double)TR::get(x, i);
+        const double r = (double)TR::get(y, i) - (a + b * xi);
         sr += r;
         srx += r * xi;
         sr2 += r * r;
     }
 
     __shared__ double lds[4 * 3];
-    double acc[3] = {(double)sr2, (double)sr, (double)srx};
+    double acc[3] = {sr2, sr, srx};
     block_reduce_add<3>(acc, lds);
     if (threadIdx.x == 0) {
         // logp = const - sum(r^2)/(2 sig^2); grads scaled by 1/sig^2

@@ -68,41 +68,57 @@ class FederatedShardEngine:
         self._distributed = use_distributed
         self._buf: Optional[torch.Tensor] = None
         self._grad_shapes: Optional[List[torch.Size]] = None
+        self._grad_views: Optional[List[torch.Tensor]] = None
 
     # -- helpers --------------------------------------------------------
-    def _ensure_buffer(self, logp: torch.Tensor, grads: Sequence[torch.Tensor]) -> torch.Tensor:
-        if self._buf is None:
-            total = 1 + sum(g.numel() for g in grads)
-            self._buf = torch.empty(total, dtype=torch.float64, device=logp.device)
-            self._grad_shapes = [g.shape for g in grads]
-        return self._buf
-
-    # -- evaluation -----------------------------------------------------
-    def logp_grad(self, *params) -> Tuple[torch.Tensor, List[torch.Tensor]]:
-        """Shard-local eval + all-reduce; returns federated (logp, grads)."""
-        import torch.distributed as dist
-
-        logp, grads = self.model.logp_grad(*params)
-        buf = self._ensure_buffer(logp, grads)
-        # Fuse into the persistent buffer.  When the model's kernel already
-        # wrote views of one contiguous fp64 buffer this is a device-side
-        # copy of a few doubles; eager paths pay one small pack.
-        buf[0] = logp.to(torch.float64)
-        off = 1
-        for g in grads:
-            n = g.numel()
-            buf[off : off + n] = g.reshape(-1).to(torch.float64)
-            off += n
-        if self._distributed:
-            dist.all_reduce(buf, op=dist.ReduceOp.SUM, group=self._group)
-        logp_total = buf[0]
-        out_grads = []
+    def _init_buffer(self, logp: torch.Tensor, grads: Sequence[torch.Tensor]) -> None:
+        total = 1 + sum(g.numel() for g in grads)
+        self._buf = torch.empty(total, dtype=torch.float64, device=logp.device)
+        self._grad_shapes = [g.shape for g in grads]
+        self._grad_views = []
         off = 1
         for shape in self._grad_shapes:
             n = int(np.prod(shape)) if len(shape) else 1
-            out_grads.append(buf[off : off + n].reshape(shape))
+            self._grad_views.append(self._buf[off : off + n].reshape(shape))
             off += n
-        return logp_total, out_grads
+
+    # -- evaluation -----------------------------------------------------
+    def logp_grad(self, *params) -> Tuple[torch.Tensor, List[torch.Tensor]]:
+        """Shard-local eval + all-reduce; returns federated (logp, grads).
+
+        The model writes its fused [logp, *grads] directly into the
+        persistent all-reduce buffer when it supports ``out=`` (the HIP
+        kernels do), so the hot path is: one kernel launch -> one
+        ``all_reduce`` -> views.  No per-call allocations or packing.
+        """
+        import torch.distributed as dist
+
+        if self._buf is None:
+            logp, grads = self.model.logp_grad(*params)
+            self._init_buffer(logp, grads)
+            self._buf[0] = logp.to(torch.float64)
+            for view, g in zip(self._grad_views, grads):
+                view.copy_(g.reshape(view.shape).to(torch.float64))
+        else:
+            try:
+                logp, grads = self.model.logp_grad(*params, out=self._buf)
+                aliased = logp.data_ptr() == self._buf.data_ptr()
+            except TypeError:  # model without out= support
+                logp, grads = self.model.logp_grad(*params)
+                aliased = False
+            if not aliased:
+                self._buf[0] = logp.to(torch.float64)
+                for view, g in zip(self._grad_views, grads):
+                    view.copy_(g.reshape(view.shape).to(torch.float64))
+        if self._distributed:
+            dist.all_reduce(self._buf, op=dist.ReduceOp.SUM, group=self._group)
+        return self._buf[0], list(self._grad_views)
+
+    def logp_grad_fused(self, *params) -> torch.Tensor:
+        """Hot-path variant returning the raw fused fp64 buffer
+        ``[logp, *grads]`` (still on device, no host sync)."""
+        self.logp_grad(*params)
+        return self._buf
 
     # -- numpy edge (what rank 0's gRPC service serves) ------------------
     def __call__(self, *params) -> Tuple[np.ndarray, List[np.ndarray]]:
