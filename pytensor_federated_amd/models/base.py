@@ -42,7 +42,8 @@ class LogpGradModel:
         """Numpy-edge evaluation (the LogpGradFunc signature)."""
         if self._delay is not None:
             t_start = time.perf_counter()
-        tparams = [torch.as_tensor(np.asarray(p, dtype=np.float64)) for p in params]
+        # copy: wire-decoded arrays are read-only zero-copy views
+        tparams = [torch.from_numpy(np.array(p, dtype=np.float64)) for p in params]
         logp, grads = self.logp_grad(*tparams)
         result = (
             np.asarray(logp.detach().cpu().double().numpy()),

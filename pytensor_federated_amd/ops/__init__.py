@@ -56,7 +56,8 @@ def _try_load() -> Optional[ctypes.CDLL]:
     lib.fed_gaussian_linear.argtypes = [
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_longlong,
         ctypes.c_double, ctypes.c_double, ctypes.c_double,
-        ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_longlong,
+        ctypes.c_int, ctypes.c_void_p,
     ]
     lib.fed_logistic_glm.restype = ctypes.c_int
     lib.fed_logistic_glm.argtypes = [
@@ -116,16 +117,27 @@ def gaussian_linear_logp_grad(
     assert x.is_cuda and y.is_cuda and x.is_contiguous() and y.is_contiguous()
     if out is None:
         out = torch.empty(3, dtype=torch.float64, device=x.device)
+    ws = _workspace(x.device, "gaussian", 3 * 2048)  # fp64[3] per block
     rc = lib.fed_gaussian_linear(
         x.data_ptr(), y.data_ptr(), x.numel(),
         float(a), float(b), float(sigma),
-        out.data_ptr(), _DTYPE_CODE[x.dtype], _stream_ptr(),
+        out.data_ptr(), ws.data_ptr(), ws.numel() * 8,
+        _DTYPE_CODE[x.dtype], _stream_ptr(),
     )
     _check(rc, "fed_gaussian_linear")
     return out[0], out[1], out[2]
 
 
 _ws_cache = {}
+
+
+def _workspace(device, kind: str, n_f64_or_f32: int, dtype=torch.float64) -> torch.Tensor:
+    key = (device.index, kind)
+    ws = _ws_cache.get(key)
+    if ws is None or ws.numel() < n_f64_or_f32:
+        ws = torch.empty(n_f64_or_f32, dtype=dtype, device=device)
+        _ws_cache[key] = ws
+    return ws
 
 
 def logistic_glm_logp_grad(
@@ -143,11 +155,7 @@ def logistic_glm_logp_grad(
     beta_f32 = beta.detach().to(device=X.device, dtype=torch.float32).contiguous()
     if out is None:
         out = torch.empty(1 + K, dtype=torch.float64, device=X.device)
-    key = (X.device.index, K)
-    ws = _ws_cache.get(key)
-    if ws is None:
-        ws = torch.empty(1024 * K, dtype=torch.float32, device=X.device)
-        _ws_cache[key] = ws
+    ws = _workspace(X.device, f"logistic{K}", 1024 * K, dtype=torch.float32)
     rc = lib.fed_logistic_glm(
         X.data_ptr(), y.data_ptr(), n, K,
         beta_f32.data_ptr(), out.data_ptr(), ws.data_ptr(),

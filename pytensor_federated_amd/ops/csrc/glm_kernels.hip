@@ -176,13 +176,36 @@ __global__ __launch_bounds__(256) void k_gaussian_linear(
     __shared__ double lds[4 * 3];
     double acc[3] = {sr2, sr, srx};
     block_reduce_add<3>(acc, lds);
+    // One plain f64x3 store per block into the partial slab.  (A fp64
+    // atomicAdd finish serializes 2048 RMWs on one address ~= 75 us --
+    // measured; the slab + tiny second kernel costs ~4 us total.)
     if (threadIdx.x == 0) {
-        // logp = const - sum(r^2)/(2 sig^2); grads scaled by 1/sig^2
-        double logp_part = -0.5 * inv_sig2 * acc[0];
-        if (blockIdx.x == 0) logp_part += logp_const;
-        atomicAdd(&out3[0], logp_part);
-        atomicAdd(&out3[1], inv_sig2 * acc[1]);
-        atomicAdd(&out3[2], inv_sig2 * acc[2]);
+        double* slab = out3 + 3 * (long long)blockIdx.x;
+        slab[0] = acc[0];
+        slab[1] = acc[1];
+        slab[2] = acc[2];
+    }
+}
+
+// Reduce the [nblocks][3] slab into out3 = {logp, d/da, d/db}. One block.
+__global__ __launch_bounds__(256) void k_gaussian_finish(
+    const double* __restrict__ slab,
+    int n_slabs,
+    double inv_sig2,
+    double logp_const,
+    double* __restrict__ out3
+) {
+    double acc[3] = {0.0, 0.0, 0.0};
+    for (int i = threadIdx.x; i < n_slabs; i += blockDim.x) {
+#pragma unroll
+        for (int k = 0; k < 3; ++k) acc[k] += slab[3 * (long long)i + k];
+    }
+    __shared__ double lds[4 * 3];
+    block_reduce_add<3>(acc, lds);
+    if (threadIdx.x == 0) {
+        out3[0] = logp_const - 0.5 * inv_sig2 * acc[0];
+        out3[1] = inv_sig2 * acc[1];
+        out3[2] = inv_sig2 * acc[2];
     }
 }
 
@@ -286,17 +309,26 @@ __global__ __launch_bounds__(256) void k_logistic_glm_reg(
 }
 
 // Column-wise fp64 reduction of the grad slab into out[1..K].
+// grid = (ceil(K/256), CHUNKS): blockIdx.y sums a slice of the slab rows and
+// fp64-atomicAdds its column partials (CHUNKS atomics per address -- cheap;
+// a 1-d grid is latency-bound: 4 blocks reading 8 MB measured 363 us).
 __global__ __launch_bounds__(256) void k_colsum_reduce(
     const float* __restrict__ slab,  // [n_slabs][K]
     int n_slabs,
     int K,
-    double* __restrict__ out_grad  // [K]
+    double* __restrict__ out_grad  // [K], pre-zeroed
 ) {
     const int col = blockIdx.x * blockDim.x + threadIdx.x;
     if (col >= K) return;
+    const int chunk = (n_slabs + gridDim.y - 1) / gridDim.y;
+    const int b0 = blockIdx.y * chunk;
+    const int b1 = min(b0 + chunk, n_slabs);
     double s = 0.0;
-    for (int b = 0; b < n_slabs; ++b) s += (double)slab[(long long)b * K + col];
-    out_grad[col] = s;
+    for (int b = b0; b < b1; ++b) s += (double)slab[(long long)b * K + col];
+    if (gridDim.y == 1)
+        out_grad[col] = s;
+    else
+        atomicAdd(&out_grad[col], s);
 }
 
 // ---------------------------------------------------------------------------
@@ -316,39 +348,45 @@ extern "C" {
 
 const char* fed_last_hip_error(void) { return hipGetErrorString(hipGetLastError()); }
 
+// workspace: >= 3 * grid * sizeof(double) (grid <= 2048 -> 48 KB)
 int fed_gaussian_linear(
     const void* x, const void* y, long long n,
     double a, double b, double sigma,
-    double* out3, int dtype, void* stream_v
+    double* out3, double* workspace, long long ws_bytes,
+    int dtype, void* stream_v
 ) {
     hipStream_t stream = (hipStream_t)stream_v;
-    hipError_t err = hipMemsetAsync(out3, 0, 3 * sizeof(double), stream);
-    if (err != hipSuccess) return (int)err;
     const double inv_sig2 = 1.0 / (sigma * sigma);
     const double logp_const = -0.5 * (double)n * log(2.0 * M_PI * sigma * sigma);
     const int block = 256;
+    int grid;
     switch (dtype) {
-        case FED_F32: {
-            const int grid = pick_grid(n / 4, block);
-            hipLaunchKernelGGL(k_gaussian_linear<float>, dim3(grid), dim3(block), 0, stream,
-                               (const float*)x, (const float*)y, n, a, b, inv_sig2, logp_const, out3);
-            break;
-        }
-        case FED_F64: {
-            const int grid = pick_grid(n / 2, block);
-            hipLaunchKernelGGL(k_gaussian_linear<double>, dim3(grid), dim3(block), 0, stream,
-                               (const double*)x, (const double*)y, n, a, b, inv_sig2, logp_const, out3);
-            break;
-        }
-        case FED_BF16: {
-            const int grid = pick_grid(n / 8, block);
-            hipLaunchKernelGGL(k_gaussian_linear<bf16_tag>, dim3(grid), dim3(block), 0, stream,
-                               (const bf16_tag*)x, (const bf16_tag*)y, n, a, b, inv_sig2, logp_const, out3);
-            break;
-        }
-        default:
-            return -2;
+        case FED_F32: grid = pick_grid(n / 4, block); break;
+        case FED_F64: grid = pick_grid(n / 2, block); break;
+        case FED_BF16: grid = pick_grid(n / 8, block); break;
+        default: return -2;
     }
+    if ((long long)grid * 3 * sizeof(double) > ws_bytes)
+        grid = (int)(ws_bytes / (3 * sizeof(double)));
+    if (grid < 1) return -3;
+    switch (dtype) {
+        case FED_F32:
+            hipLaunchKernelGGL(k_gaussian_linear<float>, dim3(grid), dim3(block), 0, stream,
+                               (const float*)x, (const float*)y, n, a, b, inv_sig2, logp_const, workspace);
+            break;
+        case FED_F64:
+            hipLaunchKernelGGL(k_gaussian_linear<double>, dim3(grid), dim3(block), 0, stream,
+                               (const double*)x, (const double*)y, n, a, b, inv_sig2, logp_const, workspace);
+            break;
+        case FED_BF16:
+            hipLaunchKernelGGL(k_gaussian_linear<bf16_tag>, dim3(grid), dim3(block), 0, stream,
+                               (const bf16_tag*)x, (const bf16_tag*)y, n, a, b, inv_sig2, logp_const, workspace);
+            break;
+    }
+    hipError_t kerr = hipGetLastError();
+    if (kerr != hipSuccess) return (int)kerr;
+    hipLaunchKernelGGL(k_gaussian_finish, dim3(1), dim3(256), 0, stream,
+                       workspace, grid, inv_sig2, logp_const, out3);
     return (int)hipGetLastError();
 }
 
@@ -399,7 +437,10 @@ int fed_logistic_glm(
     if (kerr != hipSuccess) return (int)kerr;
 
     const int rgrid = (K + 255) / 256;
-    hipLaunchKernelGGL(k_colsum_reduce, dim3(rgrid), dim3(256), 0, stream,
+    int chunks = grid / 8;
+    if (chunks < 1) chunks = 1;
+    if (chunks > 64) chunks = 64;
+    hipLaunchKernelGGL(k_colsum_reduce, dim3(rgrid, chunks), dim3(256), 0, stream,
                        workspace, grid, K, out + 1);
     return (int)hipGetLastError();
 }
