@@ -109,11 +109,21 @@ def main():
     theta_dev = torch.as_tensor(theta0, dtype=torch.float64, device=device)
     host_buf = None
 
+    # single-GPU linear serving path: one native sync call per evaluation
+    fast_sync = (
+        not distributed
+        and readback
+        and args.model == "linear"
+        and hasattr(model, "logp_grad_sync")
+    )
+
     def one_step(t: int):
         nonlocal host_buf
         # every rank derives the same perturbed theta (the broadcast of theta
         # from the driver is folded into the all-reduce round trip below)
         scale = 1.0 + 0.001 * math.sin(t)
+        if fast_sync:
+            return model.logp_grad_sync(a0 * scale, b0 * scale)
         if args.model == "linear":
             buf = engine.logp_grad_fused(a0 * scale, b0 * scale)
         else:

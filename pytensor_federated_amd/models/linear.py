@@ -135,6 +135,18 @@ class GaussianLinearModel(LogpGradModel):
             return out[0], [out[1], out[2]]
         return logp, grads
 
+    def logp_grad_sync(self, intercept, slope) -> Tuple[float, float, float]:
+        """Lowest-latency single-GPU path: one native call returning host
+        floats (kernel launch + GPU-written pinned mailbox + sync)."""
+        if self._kernel_path():
+            from ..ops import gaussian_linear_eval_sync
+
+            return gaussian_linear_eval_sync(
+                self._x, self._y, float(intercept), float(slope), self._sigma
+            )
+        logp, (ga, gb) = self._logp_grad_eager(float(intercept), float(slope))
+        return float(logp), float(ga), float(gb)
+
     def _logp_grad_eager(self, a: float, b: float) -> Tuple[torch.Tensor, List[torch.Tensor]]:
         x, y = self._x, self._y
         acc_dtype = torch.float64 if x.dtype == torch.float64 else torch.float32

@@ -19,6 +19,7 @@ __all__ = [
     "kernels_available",
     "require_kernels",
     "gaussian_linear_logp_grad",
+    "gaussian_linear_eval_sync",
     "logistic_glm_logp_grad",
 ]
 
@@ -59,6 +60,16 @@ def _try_load() -> Optional[ctypes.CDLL]:
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_longlong,
         ctypes.c_int, ctypes.c_void_p,
     ]
+    lib.fed_gaussian_linear_eval.restype = ctypes.c_int
+    lib.fed_gaussian_linear_eval.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_longlong,
+        ctypes.c_double, ctypes.c_double, ctypes.c_double,
+        ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.c_longlong,
+        ctypes.c_int, ctypes.c_void_p,
+    ]
+    lib.fed_host_alloc.restype = ctypes.c_void_p
+    lib.fed_host_alloc.argtypes = [ctypes.c_longlong]
     lib.fed_logistic_glm.restype = ctypes.c_int
     lib.fed_logistic_glm.argtypes = [
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_longlong, ctypes.c_int,
@@ -128,7 +139,59 @@ def gaussian_linear_logp_grad(
     return out[0], out[1], out[2]
 
 
+import threading
+
+import numpy as _np
+
 _ws_cache = {}
+_mailbox_lock = threading.Lock()
+_mailbox = None  # (np.ndarray view over pinned mapped memory)
+
+
+def _get_mailbox() -> "_np.ndarray":
+    """Process-wide 3-double pinned mailbox the finish kernel writes into."""
+    global _mailbox
+    if _mailbox is None:
+        lib = require_kernels()
+        ptr = lib.fed_host_alloc(3 * 8)
+        if not ptr:
+            raise RuntimeError("hipHostMalloc failed for the result mailbox")
+        _mailbox = _np.ctypeslib.as_array(
+            ctypes.cast(ptr, ctypes.POINTER(ctypes.c_double)), shape=(3,)
+        )
+    return _mailbox
+
+
+def gaussian_linear_eval_sync(
+    x: torch.Tensor,
+    y: torch.Tensor,
+    a: float,
+    b: float,
+    sigma: float,
+    out: Optional[torch.Tensor] = None,
+) -> Tuple[float, float, float]:
+    """Single-GPU hot path: ONE ctypes call = launch + GPU-written pinned
+    mailbox + stream sync; returns (logp, d/da, d/db) as host floats.
+
+    ~2x lower per-call latency than the async wrapper + torch readback
+    (no torch copy op, no separate D2H kernel, no python sync call).
+    """
+    lib = require_kernels()
+    assert x.is_cuda and x.is_contiguous() and y.is_contiguous()
+    ws = _workspace(x.device, "gaussian", 3 * 2048)
+    if out is None:
+        out = _workspace(x.device, "gaussian_out", 3)
+    with _mailbox_lock:
+        mailbox = _get_mailbox()
+        rc = lib.fed_gaussian_linear_eval(
+            x.data_ptr(), y.data_ptr(), x.numel(),
+            float(a), float(b), float(sigma),
+            out.data_ptr(), mailbox.ctypes.data,
+            ws.data_ptr(), ws.numel() * 8,
+            _DTYPE_CODE[x.dtype], _stream_ptr(),
+        )
+        _check(rc, "fed_gaussian_linear_eval")
+        return float(mailbox[0]), float(mailbox[1]), float(mailbox[2])
 
 
 def _workspace(device, kind: str, n_f64_or_f32: int, dtype=torch.float64) -> torch.Tensor:

@@ -138,3 +138,15 @@ def test_engine_single_rank_gpu(dev):
     logp, grads = engine(1.0, 0.5)
     logp_ref, grads_ref = model(1.0, 0.5)
     np.testing.assert_allclose(logp, logp_ref, rtol=1e-12)
+
+
+def test_gaussian_sync_eval_matches_async(dev):
+    from pytensor_federated_amd.ops import gaussian_linear_eval_sync
+
+    x, y = generate_linear_dataset(1_000_000, seed=31)
+    m = GaussianLinearModel(x, y, sigma=0.4, device=dev, dtype=torch.bfloat16, use_kernels=True)
+    logp_a, (ga_a, gb_a) = m(1.2, 0.6)
+    logp_s, ga_s, gb_s = m.logp_grad_sync(1.2, 0.6)
+    np.testing.assert_allclose(logp_s, logp_a, rtol=1e-12)
+    np.testing.assert_allclose(ga_s, ga_a, rtol=1e-12)
+    np.testing.assert_allclose(gb_s, gb_a, rtol=1e-12)
