@@ -93,6 +93,7 @@ class GaussianLinearModel(LogpGradModel):
         self._sigma = float(sigma)
         self._use_kernels = use_kernels
         self._n = self._x.numel()
+        self._ws = None  # per-model kernel workspace (slab + arrival ticket)
         if self._x.shape != self._y.shape or self._x.dim() != 1:
             raise ValueError("x and y must be equal-length 1-d arrays.")
 
@@ -124,7 +125,7 @@ class GaussianLinearModel(LogpGradModel):
             from ..ops import gaussian_linear_logp_grad
 
             logp, ga, gb = gaussian_linear_logp_grad(
-                self._x, self._y, a, b, self._sigma, out=out
+                self._x, self._y, a, b, self._sigma, out=out, ws=self._kernel_ws()
             )
             return logp, [ga, gb]
         logp, grads = self._logp_grad_eager(a, b)
@@ -135,14 +136,22 @@ class GaussianLinearModel(LogpGradModel):
             return out[0], [out[1], out[2]]
         return logp, grads
 
+    def _kernel_ws(self):
+        if self._ws is None:
+            from ..ops import gaussian_workspace
+
+            self._ws = gaussian_workspace(self._x.device)
+        return self._ws
+
     def logp_grad_sync(self, intercept, slope) -> Tuple[float, float, float]:
         """Lowest-latency single-GPU path: one native call returning host
-        floats (kernel launch + GPU-written pinned mailbox + sync)."""
+        floats (fused kernel + GPU-written pinned mailbox poll)."""
         if self._kernel_path():
             from ..ops import gaussian_linear_eval_sync
 
             return gaussian_linear_eval_sync(
-                self._x, self._y, float(intercept), float(slope), self._sigma
+                self._x, self._y, float(intercept), float(slope), self._sigma,
+                ws=self._kernel_ws(),
             )
         logp, (ga, gb) = self._logp_grad_eager(float(intercept), float(slope))
         return float(logp), float(ga), float(gb)

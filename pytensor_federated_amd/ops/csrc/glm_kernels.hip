@@ -125,24 +125,31 @@ __device__ inline void block_reduce_add(double* vals, double* lds /* [4][NACC] *
 // Gaussian linear regression: fused logp + d/da + d/db
 // ---------------------------------------------------------------------------
 
+// Single-launch variant: pass1 + in-launch combine by the LAST-arriving
+// block (agent-scope release/acquire per cdna_hip_programming.md §6 G16;
+// saves the finish launch + its ~4.4 us single-block latency).  The ticket
+// counter is MONOTONIC: last-of-this-call iff old % gridDim == gridDim-1,
+// so no per-call zeroing is needed (grid is a power of two -> the u32 wrap
+// stays consistent).  One workspace+ticket per model instance; calls on one
+// stream serialize, so slab reuse across calls is safe.
 template <typename T>
-__global__ __launch_bounds__(256) void k_gaussian_linear(
+__global__ __launch_bounds__(256) void k_gaussian_linear_fused(
     const T* __restrict__ x,
     const T* __restrict__ y,
     long long n,
     double a_d,
     double b_d,
     double inv_sig2,
-    double logp_const,   // -n/2 * log(2*pi*sigma^2)
-    double* __restrict__ out3  // {logp, dlogp/da, dlogp/db}, pre-zeroed
+    double logp_const,
+    double* __restrict__ slab,       // [gridDim][3]
+    unsigned* __restrict__ ticket,   // monotonic arrival counter
+    double* __restrict__ out3,       // device result
+    double* __restrict__ out3_host,  // mapped pinned mailbox (nullable)
+    unsigned long long seq           // call sequence for the mailbox flag
 ) {
     using TR = VecTraits<T>;
     using A = typename TR::acc_t;
     constexpr int VEC = TR::VEC;
-    // Element math + accumulation in fp64 for every input dtype: the kernel
-    // is HBM-bound (4 B/row bf16), so the extra f64 FMAs are free, and the
-    // cancellation-heavy sum(r) would otherwise carry a systematic fp32
-    // pred-rounding bias (~1e-3 relative at N=1e7 -- measured).
     const double a = a_d;
     const double b = b_d;
 
@@ -164,7 +171,6 @@ __global__ __launch_bounds__(256) void k_gaussian_linear(
             sr2 += r * r;
         }
     }
-    // scalar tail
     for (long long i = nvec * VEC + gid; i < n; i += gstride) {
         const double xi = (double)TR::get(x, i);
         const double r = (double)TR::get(y, i) - (a + b * xi);
@@ -173,44 +179,41 @@ __global__ __launch_bounds__(256) void k_gaussian_linear(
         sr2 += r * r;
     }
 
-    __shared__ double lds[4 * 3];
+    __shared__ double lds[4 * 3 + 1];  // ONE shared object (reduce + flag)
     double acc[3] = {sr2, sr, srx};
     block_reduce_add<3>(acc, lds);
-    // One plain f64x3 store per block into the partial slab.  (A fp64
-    // atomicAdd finish serializes 2048 RMWs on one address ~= 75 us --
-    // measured; the slab + tiny second kernel costs ~4 us total.)
     if (threadIdx.x == 0) {
-        double* slab = out3 + 3 * (long long)blockIdx.x;
-        slab[0] = acc[0];
-        slab[1] = acc[1];
-        slab[2] = acc[2];
+        double* s = slab + 3 * (long long)blockIdx.x;
+        s[0] = acc[0];
+        s[1] = acc[1];
+        s[2] = acc[2];
+        // G16 R1 publish: drain this wave's stores, agent release, re-drain
+        // (the compiler may drop the post-wbl2 wait -- pitfall 12), ticket.
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        const unsigned old =
+            __hip_atomic_fetch_add(ticket, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        const bool last = (old % gridDim.x) == (gridDim.x - 1);
+        if (last) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+        lds[12] = last ? 1.0 : 0.0;
     }
-}
+    __syncthreads();
+    if (lds[12] == 0.0) return;
 
-// Reduce the [nblocks][3] slab into out3 = {logp, d/da, d/db}. One block.
-// out3_host (optional, hipHostMalloc'd mapped memory) gets the same three
-// doubles written directly from the GPU -- the host reads them after a
-// stream sync with no separate D2H copy (saves one launch + ~4 us).
-__global__ __launch_bounds__(256) void k_gaussian_finish(
-    const double* __restrict__ slab,
-    int n_slabs,
-    double inv_sig2,
-    double logp_const,
-    double* __restrict__ out3,
-    double* __restrict__ out3_host
-) {
-    double acc[3] = {0.0, 0.0, 0.0};
-    // 24 B per slab entry: lane-parallel over entries, sequential k
-    for (int i = threadIdx.x; i < n_slabs; i += blockDim.x) {
+    // last-arriving block: reduce every block's slab entry (plain loads are
+    // valid after the acquire + barrier above)
+    double fin[3] = {0.0, 0.0, 0.0};
+    for (unsigned i = threadIdx.x; i < gridDim.x; i += blockDim.x) {
 #pragma unroll
-        for (int k = 0; k < 3; ++k) acc[k] += slab[3 * (long long)i + k];
+        for (int k = 0; k < 3; ++k) fin[k] += slab[3 * (long long)i + k];
     }
-    __shared__ double lds[4 * 3];
-    block_reduce_add<3>(acc, lds);
+    __syncthreads();  // lds[0..11] reused below
+    block_reduce_add<3>(fin, lds);
     if (threadIdx.x == 0) {
-        const double v0 = logp_const - 0.5 * inv_sig2 * acc[0];
-        const double v1 = inv_sig2 * acc[1];
-        const double v2 = inv_sig2 * acc[2];
+        const double v0 = logp_const - 0.5 * inv_sig2 * fin[0];
+        const double v1 = inv_sig2 * fin[1];
+        const double v2 = inv_sig2 * fin[2];
         out3[0] = v0;
         out3[1] = v1;
         out3[2] = v2;
@@ -218,6 +221,8 @@ __global__ __launch_bounds__(256) void k_gaussian_finish(
             out3_host[0] = v0;
             out3_host[1] = v1;
             out3_host[2] = v2;
+            __threadfence_system();  // results visible to host before the flag
+            ((unsigned long long*)out3_host)[3] = seq;
         }
     }
 }
@@ -268,7 +273,48 @@ __global__ __launch_bounds__(256) void k_logistic_glm_reg(
         }
 
     double logp_acc = 0.0;
-    for (long long r = wave_id; r < n_rows; r += n_waves) {
+    // Two rows per iteration: both rows' global loads issue before either
+    // row's dependent dot/reduce/transcendental chain, so each wave keeps
+    // ~2x the HBM traffic in flight (the 1-row loop measured 68% of HBM
+    // peak; the dependent per-row chain was the gap).
+    float xreg2[KITER][VEC];
+    const long long pair_stride = n_waves * 2;
+    long long r = wave_id * 2;
+    for (; r + 1 < n_rows; r += pair_stride) {
+        const T* row0 = X + r * (long long)K;
+        const T* row1 = row0 + K;
+        float z0p = 0.f, z1p = 0.f;
+#pragma unroll
+        for (int c = 0; c < KITER; ++c) {
+            TR::load(row0 + c * WAVE * VEC + lane * VEC, xreg[c]);
+            TR::load(row1 + c * WAVE * VEC + lane * VEC, xreg2[c]);
+        }
+#pragma unroll
+        for (int c = 0; c < KITER; ++c)
+#pragma unroll
+            for (int j = 0; j < VEC; ++j) {
+                z0p += xreg[c][j] * breg[c][j];
+                z1p += xreg2[c][j] * breg[c][j];
+            }
+        float z0 = wave_reduce_sum_f32(z0p);
+        float z1 = wave_reduce_sum_f32(z1p);
+        z0 = __shfl(z0, 0, WAVE);
+        z1 = __shfl(z1, 0, WAVE);
+        const float y0 = TR::get(y, r);
+        const float y1 = TR::get(y, r + 1);
+        // stable: y*z - softplus(z) = y*z - (max(z,0) + log1p(exp(-|z|)))
+        const float sp0 = fmaxf(z0, 0.f) + log1pf(__expf(-fabsf(z0)));
+        const float sp1 = fmaxf(z1, 0.f) + log1pf(__expf(-fabsf(z1)));
+        if (lane == 0) logp_acc += (double)(y0 * z0 - sp0) + (double)(y1 * z1 - sp1);
+        const float res0 = y0 - 1.f / (1.f + __expf(-z0));
+        const float res1 = y1 - 1.f / (1.f + __expf(-z1));
+#pragma unroll
+        for (int c = 0; c < KITER; ++c)
+#pragma unroll
+            for (int j = 0; j < VEC; ++j)
+                gacc[c][j] += res0 * xreg[c][j] + res1 * xreg2[c][j];
+    }
+    for (; r < n_rows; r += pair_stride) {  // odd tail row of this wave
         const T* row = X + r * (long long)K;
         float z_part = 0.f;
 #pragma unroll
@@ -278,9 +324,8 @@ __global__ __launch_bounds__(256) void k_logistic_glm_reg(
             for (int j = 0; j < VEC; ++j) z_part += xreg[c][j] * breg[c][j];
         }
         float z = wave_reduce_sum_f32(z_part);
-        z = __shfl(z, 0, WAVE);  // broadcast the row dot to all lanes
+        z = __shfl(z, 0, WAVE);
         const float yr = TR::get(y, r);
-        // stable: y*z - softplus(z) = y*z - (max(z,0) + log1p(exp(-|z|)))
         const float sp = fmaxf(z, 0.f) + log1pf(__expf(-fabsf(z)));
         if (lane == 0) logp_acc += (double)(yr * z - sp);
         const float resid = yr - 1.f / (1.f + __expf(-z));
@@ -361,13 +406,14 @@ extern "C" {
 
 const char* fed_last_hip_error(void) { return hipGetErrorString(hipGetLastError()); }
 
-// workspace: >= 3 * grid * sizeof(double) (grid <= 2048 -> 48 KB)
+// workspace layout: [0] = monotonic ticket (MUST be zero-initialized once),
+// [1..1+3*grid) = fp64 partial slab.  ws_bytes >= (1 + 3*grid_max) * 8.
 static int gaussian_linear_impl(
     const void* x, const void* y, long long n,
     double a, double b, double sigma,
     double* out3, double* out3_host,
     double* workspace, long long ws_bytes,
-    int dtype, hipStream_t stream
+    int dtype, hipStream_t stream, unsigned long long seq
 ) {
     const double inv_sig2 = 1.0 / (sigma * sigma);
     const double logp_const = -0.5 * (double)n * log(2.0 * M_PI * sigma * sigma);
@@ -379,27 +425,30 @@ static int gaussian_linear_impl(
         case FED_BF16: grid = pick_grid(n / 8, block); break;
         default: return -2;
     }
-    if ((long long)grid * 3 * sizeof(double) > ws_bytes)
-        grid = (int)(ws_bytes / (3 * sizeof(double)));
+    long long ws_cap = (ws_bytes / 8 - 1) / 3;
+    if (grid > ws_cap) grid = (int)ws_cap;
     if (grid < 1) return -3;
+    // power of two so the monotonic u32 ticket wraps consistently
+    while (grid & (grid - 1)) grid &= grid - 1;
+    unsigned* ticket = (unsigned*)workspace;
+    double* slab = workspace + 1;
     switch (dtype) {
         case FED_F32:
-            hipLaunchKernelGGL(k_gaussian_linear<float>, dim3(grid), dim3(block), 0, stream,
-                               (const float*)x, (const float*)y, n, a, b, inv_sig2, logp_const, workspace);
+            hipLaunchKernelGGL(k_gaussian_linear_fused<float>, dim3(grid), dim3(block), 0, stream,
+                               (const float*)x, (const float*)y, n, a, b, inv_sig2, logp_const,
+                               slab, ticket, out3, out3_host, seq);
             break;
         case FED_F64:
-            hipLaunchKernelGGL(k_gaussian_linear<double>, dim3(grid), dim3(block), 0, stream,
-                               (const double*)x, (const double*)y, n, a, b, inv_sig2, logp_const, workspace);
+            hipLaunchKernelGGL(k_gaussian_linear_fused<double>, dim3(grid), dim3(block), 0, stream,
+                               (const double*)x, (const double*)y, n, a, b, inv_sig2, logp_const,
+                               slab, ticket, out3, out3_host, seq);
             break;
         case FED_BF16:
-            hipLaunchKernelGGL(k_gaussian_linear<bf16_tag>, dim3(grid), dim3(block), 0, stream,
-                               (const bf16_tag*)x, (const bf16_tag*)y, n, a, b, inv_sig2, logp_const, workspace);
+            hipLaunchKernelGGL(k_gaussian_linear_fused<bf16_tag>, dim3(grid), dim3(block), 0, stream,
+                               (const bf16_tag*)x, (const bf16_tag*)y, n, a, b, inv_sig2, logp_const,
+                               slab, ticket, out3, out3_host, seq);
             break;
     }
-    hipError_t kerr = hipGetLastError();
-    if (kerr != hipSuccess) return (int)kerr;
-    hipLaunchKernelGGL(k_gaussian_finish, dim3(1), dim3(256), 0, stream,
-                       workspace, grid, inv_sig2, logp_const, out3, out3_host);
     return (int)hipGetLastError();
 }
 
@@ -410,18 +459,18 @@ int fed_gaussian_linear(
     int dtype, void* stream_v
 ) {
     return gaussian_linear_impl(x, y, n, a, b, sigma, out3, nullptr,
-                                workspace, ws_bytes, dtype, (hipStream_t)stream_v);
+                                workspace, ws_bytes, dtype, (hipStream_t)stream_v, 0);
 }
 
-// Synchronous single-call evaluation: launch + GPU writes the 3 results
-// into mapped pinned host memory + stream sync.  ONE ctypes round trip per
-// logp+grad call on the 1-GPU serving path.
+// Synchronous single-call evaluation: ONE launch; the last-arriving block
+// writes {logp, ga, gb, seq} into the mapped pinned mailbox; the host
+// spin-reads the seq flag (no hipStreamSynchronize on the happy path).
 int fed_gaussian_linear_eval(
     const void* x, const void* y, long long n,
     double a, double b, double sigma,
     double* out3_dev, double* out3_host,
     double* workspace, long long ws_bytes,
-    int dtype, void* stream_v
+    int dtype, void* stream_v, unsigned long long seq
 ) {
     hipStream_t stream = (hipStream_t)stream_v;
     void* mailbox_dev = nullptr;  // device-side alias of the pinned mailbox
@@ -429,9 +478,16 @@ int fed_gaussian_linear_eval(
     if (perr != hipSuccess) return (int)perr;
     int rc = gaussian_linear_impl(x, y, n, a, b, sigma, out3_dev,
                                   (double*)mailbox_dev, workspace, ws_bytes,
-                                  dtype, stream);
+                                  dtype, stream, seq);
     if (rc != 0) return rc;
-    return (int)hipStreamSynchronize(stream);
+    volatile unsigned long long* flag = ((volatile unsigned long long*)out3_host) + 3;
+    for (long long spins = 0; spins < 400000000LL; ++spins) {  // ~>1 s bound
+        if (*flag == seq) return 0;
+    }
+    // flag never arrived: drain the stream and surface the real error
+    hipError_t serr = hipStreamSynchronize(stream);
+    if (serr != hipSuccess) return (int)serr;
+    return (*flag == seq) ? 0 : -5;
 }
 
 // Mapped pinned host memory for the GPU-written result mailbox.

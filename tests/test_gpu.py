@@ -150,3 +150,18 @@ def test_gaussian_sync_eval_matches_async(dev):
     np.testing.assert_allclose(logp_s, logp_a, rtol=1e-12)
     np.testing.assert_allclose(ga_s, ga_a, rtol=1e-12)
     np.testing.assert_allclose(gb_s, gb_a, rtol=1e-12)
+
+
+def test_fused_combine_repeated_calls_stable(dev):
+    """Stress the in-launch last-arriver combine: 200 repeated calls must
+    all return the identical correct result (catches G16 visibility bugs --
+    the consumer block is L1-warm across calls)."""
+    x, y = generate_linear_dataset(3_000_000, seed=32)
+    m = GaussianLinearModel(x, y, sigma=0.4, device=dev, dtype=torch.bfloat16, use_kernels=True)
+    ref = m.logp_grad_sync(1.1, 0.4)
+    for i in range(200):
+        got = m.logp_grad_sync(1.1, 0.4)
+        assert got == ref, f"call {i}: {got} != {ref}"
+    # and the async path agrees
+    logp, (ga, gb) = m(1.1, 0.4)
+    np.testing.assert_allclose(float(logp), ref[0], rtol=1e-12)
