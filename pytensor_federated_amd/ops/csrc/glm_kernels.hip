@@ -96,6 +96,25 @@ __device__ __forceinline__ float wave_reduce_sum_f32(float v) {
     return v;
 }
 
+// Agent-scope write-through (sc1) accessors for cross-workgroup slabs
+// (G16 R1's cheap form: sc1 stores + drained ticket on the producer, sc1
+// loads on the consumer -- no release/acquire fences, which cost ~1.7 us
+// per block and made a fence-based combine 2.7x slower than two kernels).
+typedef __attribute__((address_space(1))) unsigned long long gu64_t;
+typedef __attribute__((address_space(1))) unsigned gu32_t;
+
+__device__ __forceinline__ void store_sc1_f64(double* p, double v) {
+    union { double d; unsigned long long u; } c;
+    c.d = v;
+    __hip_atomic_store((gu64_t*)p, c.u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+}
+
+__device__ __forceinline__ double load_sc1_f64(const double* p) {
+    union { unsigned long long u; double d; } c;
+    c.u = __hip_atomic_load((const gu64_t*)p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    return c.d;
+}
+
 // Block-level fp64 reduction of NACC per-lane values via LDS; the block's
 // totals land in lane 0 of wave 0.  BLOCK = 256 threads = 4 waves.
 template <int NACC>
@@ -184,29 +203,26 @@ __global__ __launch_bounds__(256) void k_gaussian_linear_fused(
     block_reduce_add<3>(acc, lds);
     if (threadIdx.x == 0) {
         double* s = slab + 3 * (long long)blockIdx.x;
-        s[0] = acc[0];
-        s[1] = acc[1];
-        s[2] = acc[2];
-        // G16 R1 publish: drain this wave's stores, agent release, re-drain
-        // (the compiler may drop the post-wbl2 wait -- pitfall 12), ticket.
+        // write-through publish: sc1 stores leave L2 immediately; drain this
+        // wave's stores before the ticket so the count never overtakes them
+        store_sc1_f64(&s[0], acc[0]);
+        store_sc1_f64(&s[1], acc[1]);
+        store_sc1_f64(&s[2], acc[2]);
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-        const unsigned old =
-            __hip_atomic_fetch_add(ticket, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        const unsigned old = __hip_atomic_fetch_add(
+            (gu32_t*)ticket, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
         const bool last = (old % gridDim.x) == (gridDim.x - 1);
-        if (last) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
         lds[12] = last ? 1.0 : 0.0;
     }
     __syncthreads();
     if (lds[12] == 0.0) return;
 
-    // last-arriving block: reduce every block's slab entry (plain loads are
-    // valid after the acquire + barrier above)
+    // last-arriving block: reduce every block's slab entry with sc1 loads
+    // (L1-bypassing; matches the sc1-stored data, no acquire fence needed)
     double fin[3] = {0.0, 0.0, 0.0};
     for (unsigned i = threadIdx.x; i < gridDim.x; i += blockDim.x) {
 #pragma unroll
-        for (int k = 0; k < 3; ++k) fin[k] += slab[3 * (long long)i + k];
+        for (int k = 0; k < 3; ++k) fin[k] += load_sc1_f64(&slab[3 * (long long)i + k]);
     }
     __syncthreads();  // lds[0..11] reused below
     block_reduce_add<3>(fin, lds);
