@@ -152,8 +152,8 @@ _mailbox_lock = threading.Lock()
 _mailbox = None  # (np.ndarray view over pinned mapped memory)
 _mailbox_seq = 0
 
-#: fp64 words in a gaussian workspace: 1 ticket + 3 per block (grid<=2048)
-GAUSSIAN_WS_SIZE = 1 + 3 * 2048
+#: fp64 words in a gaussian workspace: 9 spaced ticket words + 3 per block
+GAUSSIAN_WS_SIZE = 72 + 3 * 2048
 
 
 def _get_mailbox() -> "_np.ndarray":

@@ -209,9 +209,21 @@ __global__ __launch_bounds__(256) void k_gaussian_linear_fused(
         store_sc1_f64(&s[1], acc[1]);
         store_sc1_f64(&s[2], acc[2]);
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        // Sharded arrival tickets (one word saturates at ~88 atomics/us ->
+        // 2048 arrivals ~ 23 us measured; 8 group words on separate cache
+        // lines + a top word cut that ~8x).  Grouping is by blockIdx
+        // parity -- a pure partition, no XCD-placement assumption.
+        const unsigned ngroups = gridDim.x < 8 ? gridDim.x : 8;
+        const unsigned gsize = gridDim.x / ngroups;
+        const unsigned grp = blockIdx.x % ngroups;
+        bool last = false;
         const unsigned old = __hip_atomic_fetch_add(
-            (gu32_t*)ticket, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-        const bool last = (old % gridDim.x) == (gridDim.x - 1);
+            (gu32_t*)(ticket + 16 * grp), 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        if (old % gsize == gsize - 1) {  // last arrival of this group
+            const unsigned t = __hip_atomic_fetch_add(
+                (gu32_t*)(ticket + 16 * 8), 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+            last = (t % ngroups) == (ngroups - 1);
+        }
         lds[12] = last ? 1.0 : 0.0;
     }
     __syncthreads();
@@ -422,8 +434,9 @@ extern "C" {
 
 const char* fed_last_hip_error(void) { return hipGetErrorString(hipGetLastError()); }
 
-// workspace layout: [0] = monotonic ticket (MUST be zero-initialized once),
-// [1..1+3*grid) = fp64 partial slab.  ws_bytes >= (1 + 3*grid_max) * 8.
+// workspace layout (fp64 words): [0..72) = 9 ticket words, 64 B apart
+// (8 groups + 1 top; MUST be zero-initialized once), [72..72+3*grid) =
+// fp64 partial slab.  ws_bytes >= (72 + 3*grid_max) * 8.
 static int gaussian_linear_impl(
     const void* x, const void* y, long long n,
     double a, double b, double sigma,
@@ -441,13 +454,13 @@ static int gaussian_linear_impl(
         case FED_BF16: grid = pick_grid(n / 8, block); break;
         default: return -2;
     }
-    long long ws_cap = (ws_bytes / 8 - 1) / 3;
+    long long ws_cap = (ws_bytes / 8 - 72) / 3;
     if (grid > ws_cap) grid = (int)ws_cap;
     if (grid < 1) return -3;
-    // power of two so the monotonic u32 ticket wraps consistently
+    // power of two so the monotonic u32 tickets wrap consistently
     while (grid & (grid - 1)) grid &= grid - 1;
-    unsigned* ticket = (unsigned*)workspace;
-    double* slab = workspace + 1;
+    unsigned* ticket = (unsigned*)workspace;  // 9 words, 64 B apart
+    double* slab = workspace + 72;
     switch (dtype) {
         case FED_F32:
             hipLaunchKernelGGL(k_gaussian_linear_fused<float>, dim3(grid), dim3(block), 0, stream,
