@@ -423,9 +423,21 @@ __global__ __launch_bounds__(256) void k_colsum_reduce(
 
 enum FedDtype { FED_F32 = 0, FED_F64 = 1, FED_BF16 = 2 };
 
+#include <stdlib.h>
+
+static int grid_cap(void) {
+    static int cap = 0;
+    if (cap == 0) {
+        const char* env = getenv("FED_GRID_CAP");
+        cap = env ? atoi(env) : 2048;  // G11: cap + grid-stride
+        if (cap < 1 || cap > 8192) cap = 2048;
+    }
+    return cap;
+}
+
 static inline int pick_grid(long long work_items, int block) {
     long long blocks = (work_items + block - 1) / block;
-    if (blocks > 2048) blocks = 2048;  // G11: cap + grid-stride
+    if (blocks > grid_cap()) blocks = grid_cap();
     if (blocks < 1) blocks = 1;
     return (int)blocks;
 }
