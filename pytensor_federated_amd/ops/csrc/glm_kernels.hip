@@ -429,7 +429,7 @@ static int grid_cap(void) {
     static int cap = 0;
     if (cap == 0) {
         const char* env = getenv("FED_GRID_CAP");
-        cap = env ? atoi(env) : 2048;  // G11: cap + grid-stride
+        cap = env ? atoi(env) : 1024;  // G11: cap + grid-stride (A/B on MI355X: 512-1024 best)
         if (cap < 1 || cap > 8192) cap = 2048;
     }
     return cap;
