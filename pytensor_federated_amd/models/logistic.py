@@ -87,9 +87,21 @@ class LogisticGLMModel(LogpGradModel):
         return int(self._k)
 
     def _kernel_path(self) -> bool:
-        if self._use_kernels is None:
-            return self._X.is_cuda
-        return self._use_kernels
+        want = self._X.is_cuda if self._use_kernels is None else self._use_kernels
+        if want and self._use_kernels is None:
+            from ..ops import logistic_kernel_supports
+
+            if not logistic_kernel_supports(self._X.dtype, self._k):
+                import warnings
+
+                warnings.warn(
+                    f"fused logistic kernel not compiled for (dtype={self._X.dtype}, "
+                    f"K={self._k}); using the eager two-matmul path (~2x slower). "
+                    f"Supported K: multiples of 512 (bf16) / 256 (f32).",
+                    stacklevel=3,
+                )
+                return False
+        return want
 
     @property
     def fused_size(self) -> int:

@@ -224,6 +224,21 @@ def _workspace(device, kind: str, n_elems: int, dtype=torch.float64, zeroed=Fals
     return ws
 
 
+#: (dtype, K) combinations the fused logistic kernel is compiled for
+LOGISTIC_SUPPORTED = {
+    (torch.bfloat16, 512),
+    (torch.bfloat16, 1024),
+    (torch.bfloat16, 2048),
+    (torch.float32, 256),
+    (torch.float32, 512),
+    (torch.float32, 1024),
+}
+
+
+def logistic_kernel_supports(dtype, K: int) -> bool:
+    return (dtype, int(K)) in LOGISTIC_SUPPORTED
+
+
 def logistic_glm_logp_grad(
     X: torch.Tensor,
     y: torch.Tensor,

@@ -160,7 +160,29 @@ class ArraysToArraysService:
 
 
 def _gpu_load_percent() -> Optional[Tuple[float, float]]:
-    """(GPU busy %, VRAM used %) via torch/amdsmi, or None off-GPU."""
+    """(GPU busy %, HBM used %) -- the MI355X analog of the reference's
+    CPU/RAM telemetry behind the same GetLoad message (service.py:88-96).
+
+    Prefers amdsmi (the ROCm system-management library); falls back to
+    torch's utilization counters; None when no GPU is present.
+    """
+    try:
+        import amdsmi
+
+        amdsmi.amdsmi_init()
+        try:
+            handles = amdsmi.amdsmi_get_processor_handles()
+            if not handles:
+                return None
+            h = handles[0]
+            busy = float(amdsmi.amdsmi_get_gpu_activity(h)["gfx_activity"])
+            vram = amdsmi.amdsmi_get_gpu_vram_usage(h)
+            vram_pct = vram["vram_used"] / max(1, vram["vram_total"]) * 100.0
+            return busy, vram_pct
+        finally:
+            amdsmi.amdsmi_shut_down()
+    except Exception:
+        pass
     try:
         import torch
 
