@@ -1,0 +1,13 @@
+"""Self-contained inference drivers over the LogpGradFunc contract.
+
+The reference delegates sampling to PyMC (demo_model.py:38-44 runs
+``pm.find_MAP`` + ``pm.sample``).  This framework works under PyMC too (via
+``wrapper_ops``), but also ships its own drivers so a deployment needs
+nothing beyond this package: every driver consumes the framework's
+``LogpFunc`` / ``LogpGradFunc`` signatures -- a local model, a remote
+``LogpGradServiceClient.evaluate``, or a ``FederatedShardEngine`` all plug
+in unchanged (the logp+grad of one MCMC step is ONE fused worker call).
+"""
+from .map import find_map  # noqa: F401
+from .mcmc import Metropolis, sample_metropolis  # noqa: F401
+from .nuts import NUTS, sample_nuts  # noqa: F401

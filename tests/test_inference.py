@@ -1,0 +1,99 @@
+"""Inference drivers: MAP, Metropolis, NUTS against closed-form posteriors."""
+import numpy as np
+import pytest
+
+from pytensor_federated_amd.inference import find_map, sample_metropolis, sample_nuts
+from pytensor_federated_amd.models import GaussianLinearModel, generate_linear_dataset
+
+
+def gaussian_2d_logp_grad(mu, cov_inv):
+    """N(mu, cov) as a LogpGradFunc over one 2-vector parameter."""
+
+    def func(theta):
+        d = np.asarray(theta, dtype=np.float64) - mu
+        g = -cov_inv @ d
+        return np.asarray(-0.5 * d @ cov_inv @ d), [g]
+
+    return func
+
+
+class TestFindMap:
+    def test_quadratic_max(self):
+        mu = np.array([1.0, -2.0])
+        cov_inv = np.array([[2.0, 0.3], [0.3, 1.0]])
+        theta, logp = find_map(
+            gaussian_2d_logp_grad(mu, cov_inv), [np.zeros(2)], steps=800, lr=0.1
+        )
+        np.testing.assert_allclose(theta[0], mu, atol=1e-3)
+        assert logp == pytest.approx(0.0, abs=1e-5)
+
+    def test_linear_model_map_recovers_ols(self):
+        x, y = generate_linear_dataset(200, seed=41)
+        model = GaussianLinearModel(x, y, sigma=0.4)
+        theta, _ = find_map(model.as_logp_grad_func(), [np.array(0.0), np.array(0.0)],
+                            steps=2000, lr=0.05)
+        # flat-prior MAP == OLS estimate
+        A = np.stack([np.ones_like(x), x], axis=1)
+        beta_ols, *_ = np.linalg.lstsq(A, y, rcond=None)
+        np.testing.assert_allclose([float(theta[0]), float(theta[1])], beta_ols, atol=5e-3)
+
+
+class TestMetropolis:
+    def test_recovers_1d_gaussian(self):
+        def logp(theta):
+            t = float(theta)
+            return np.asarray(-0.5 * (t - 3.0) ** 2 / 4.0)
+
+        chain = sample_metropolis(
+            logp, [np.array(0.0)], draws=4000, tune=1000, seed=0
+        )
+        samples = np.array([float(d[0]) for d in chain])
+        assert abs(samples.mean() - 3.0) < 0.25
+        assert abs(samples.std() - 2.0) < 0.35
+
+
+class TestNUTS:
+    def test_recovers_correlated_gaussian(self):
+        mu = np.array([1.0, -1.0])
+        cov = np.array([[1.0, 0.6], [0.6, 2.0]])
+        cov_inv = np.linalg.inv(cov)
+        chain = sample_nuts(
+            gaussian_2d_logp_grad(mu, cov_inv),
+            [np.zeros(2)],
+            draws=1500,
+            tune=600,
+            seed=1,
+        )
+        samples = np.stack([d[0] for d in chain])
+        np.testing.assert_allclose(samples.mean(axis=0), mu, atol=0.15)
+        np.testing.assert_allclose(np.cov(samples.T), cov, atol=0.45)
+
+    def test_linear_model_posterior_matches_conjugate(self):
+        """Flat-prior linear regression: posterior = N(beta_hat, sig^2 (A^T A)^-1)."""
+        x, y = generate_linear_dataset(120, seed=42)
+        sigma = 0.4
+        model = GaussianLinearModel(x, y, sigma=sigma)
+
+        def logp_grad(theta):
+            logp, grads = model.as_logp_grad_func()(theta[0], theta[1])
+            return logp, [np.array([float(grads[0]), float(grads[1])])]
+
+        chain = sample_nuts(logp_grad, [np.zeros(2)], draws=1200, tune=600, seed=2)
+        samples = np.stack([d[0] for d in chain])
+
+        A = np.stack([np.ones_like(x), x], axis=1)
+        beta_hat, *_ = np.linalg.lstsq(A, y, rcond=None)
+        post_cov = sigma**2 * np.linalg.inv(A.T @ A)
+        post_sd = np.sqrt(np.diag(post_cov))
+        for k in range(2):
+            assert abs(samples.mean(axis=0)[k] - beta_hat[k]) < 6 * post_sd[k]
+            assert abs(samples.std(axis=0)[k] - post_sd[k]) < 0.4 * post_sd[k]
+
+    def test_divergences_rare_on_gaussian(self):
+        from pytensor_federated_amd.inference.nuts import NUTS
+
+        sampler = NUTS(gaussian_2d_logp_grad(np.zeros(2), np.eye(2)), [np.zeros(2)], seed=3)
+        for _ in range(50):
+            sampler.step()
+            sampler.adapt_step_size()
+        assert sampler.n_divergent == 0
