@@ -598,3 +598,152 @@ int fed_logistic_glm(
 }
 
 }  // extern "C"
+
+// ---------------------------------------------------------------------------
+// Native multi-shard linear engine: N shards on one GPU, one call per eval
+// ---------------------------------------------------------------------------
+//
+// The native successor of the reference's ParallelAsyncOp fan-out
+// (reference op_async.py:107-132): N federated shards resident on one
+// MI355X evaluate CONCURRENTLY, each on its own HIP stream; a combine
+// kernel (gated by per-shard events) sums the per-shard [logp, ga, gb]
+// and writes the total to the device buffer + pinned mailbox.  The whole
+// fan-out costs ONE host call; the host polls the mailbox seq flag.
+
+__global__ __launch_bounds__(64) void k_sum_shards3(
+    const double* __restrict__ shard_out,  // [n_shards][3]
+    int n_shards,
+    double* __restrict__ out3,
+    double* __restrict__ out3_host,
+    unsigned long long seq
+) {
+    if (threadIdx.x == 0) {
+        double v[3] = {0.0, 0.0, 0.0};
+        for (int s = 0; s < n_shards; ++s)
+            for (int k = 0; k < 3; ++k) v[k] += shard_out[3 * s + k];
+        out3[0] = v[0]; out3[1] = v[1]; out3[2] = v[2];
+        if (out3_host != nullptr) {
+            out3_host[0] = v[0]; out3_host[1] = v[1]; out3_host[2] = v[2];
+            __threadfence_system();
+            ((unsigned long long*)out3_host)[3] = seq;
+        }
+    }
+}
+
+#define FED_MAX_SHARDS 64
+
+struct FedLinearEngine {
+    int n_shards;
+    int dtype;
+    double sigma;
+    const void* xs[FED_MAX_SHARDS];
+    const void* ys[FED_MAX_SHARDS];
+    long long ns[FED_MAX_SHARDS];
+    hipStream_t streams[FED_MAX_SHARDS];
+    hipEvent_t events[FED_MAX_SHARDS];
+    double* ws[FED_MAX_SHARDS];      // per-shard ticket+slab workspace
+    double* shard_out;               // [n_shards][3] device
+    double* out3;                    // device total
+    double* mailbox;                 // pinned mapped {3 results, seq}
+    double* mailbox_dev;
+    unsigned long long seq;
+};
+
+extern "C" {
+
+int fed_linear_engine_destroy(void* handle);
+
+void* fed_linear_engine_create(
+    int n_shards, const void** xs, const void** ys, const long long* ns,
+    double sigma, int dtype
+) {
+    if (n_shards < 1 || n_shards > FED_MAX_SHARDS) return nullptr;
+    FedLinearEngine* e = new FedLinearEngine();
+    e->n_shards = n_shards;
+    e->dtype = dtype;
+    e->sigma = sigma;
+    e->seq = 0;
+    const long long ws_words = 72 + 3 * 2048;
+    for (int s = 0; s < n_shards; ++s) {
+        e->xs[s] = xs[s];
+        e->ys[s] = ys[s];
+        e->ns[s] = ns[s];
+        if (hipStreamCreateWithFlags(&e->streams[s], hipStreamNonBlocking) != hipSuccess ||
+            hipEventCreateWithFlags(&e->events[s], hipEventDisableTiming) != hipSuccess ||
+            hipMalloc(&e->ws[s], ws_words * 8) != hipSuccess ||
+            hipMemset(e->ws[s], 0, ws_words * 8) != hipSuccess) {
+            fed_linear_engine_destroy(e);
+            return nullptr;
+        }
+    }
+    if (hipMalloc(&e->shard_out, n_shards * 3 * 8) != hipSuccess ||
+        hipMalloc(&e->out3, 3 * 8) != hipSuccess ||
+        hipHostMalloc((void**)&e->mailbox, 4 * 8, hipHostMallocMapped) != hipSuccess ||
+        hipHostGetDevicePointer((void**)&e->mailbox_dev, e->mailbox, 0) != hipSuccess) {
+        fed_linear_engine_destroy(e);
+        return nullptr;
+    }
+    e->mailbox[3] = 0.0;
+    return e;
+}
+
+// One federated fan-out evaluation; blocks until the summed [logp, ga, gb]
+// is in out3_host_result.  sync_stream (torch current stream) is made to
+// wait for the combine so subsequent torch work stays ordered.
+int fed_linear_engine_eval(
+    void* handle, double a, double b, double* out3_host_result, void* sync_stream
+) {
+    FedLinearEngine* e = (FedLinearEngine*)handle;
+    e->seq += 1;
+    for (int s = 0; s < e->n_shards; ++s) {
+        int rc = gaussian_linear_impl(
+            e->xs[s], e->ys[s], e->ns[s], a, b, e->sigma,
+            e->shard_out + 3 * s, nullptr,
+            e->ws[s], (72 + 3 * 2048) * 8, e->dtype, e->streams[s], 0);
+        if (rc != 0) return rc;
+        if (hipEventRecord(e->events[s], e->streams[s]) != hipSuccess) return -7;
+    }
+    for (int s = 1; s < e->n_shards; ++s)
+        if (hipStreamWaitEvent(e->streams[0], e->events[s], 0) != hipSuccess) return -8;
+    hipLaunchKernelGGL(k_sum_shards3, dim3(1), dim3(64), 0, e->streams[0],
+                       e->shard_out, e->n_shards, e->out3, e->mailbox_dev, e->seq);
+    hipError_t kerr = hipGetLastError();
+    if (kerr != hipSuccess) return (int)kerr;
+    if (sync_stream != nullptr) {
+        if (hipEventRecord(e->events[0], e->streams[0]) != hipSuccess) return -7;
+        if (hipStreamWaitEvent((hipStream_t)sync_stream, e->events[0], 0) != hipSuccess)
+            return -8;
+    }
+    volatile unsigned long long* flag = ((volatile unsigned long long*)e->mailbox) + 3;
+    for (long long spins = 0; spins < 400000000LL; ++spins) {
+        if (*flag == e->seq) {
+            out3_host_result[0] = e->mailbox[0];
+            out3_host_result[1] = e->mailbox[1];
+            out3_host_result[2] = e->mailbox[2];
+            return 0;
+        }
+    }
+    hipError_t serr = hipStreamSynchronize(e->streams[0]);
+    if (serr != hipSuccess) return (int)serr;
+    if (*flag != e->seq) return -5;
+    out3_host_result[0] = e->mailbox[0];
+    out3_host_result[1] = e->mailbox[1];
+    out3_host_result[2] = e->mailbox[2];
+    return 0;
+}
+
+int fed_linear_engine_destroy(void* handle) {
+    FedLinearEngine* e = (FedLinearEngine*)handle;
+    for (int s = 0; s < e->n_shards; ++s) {
+        if (e->ws[s]) (void)hipFree(e->ws[s]);
+        if (e->streams[s]) (void)hipStreamDestroy(e->streams[s]);
+        if (e->events[s]) (void)hipEventDestroy(e->events[s]);
+    }
+    if (e->shard_out) (void)hipFree(e->shard_out);
+    if (e->out3) (void)hipFree(e->out3);
+    if (e->mailbox) (void)hipHostFree(e->mailbox);
+    delete e;
+    return 0;
+}
+
+}  // extern "C"

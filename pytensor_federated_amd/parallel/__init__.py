@@ -6,4 +6,4 @@ See SURVEY.md §2.4: the reference's parallelism axes map to
   and the asyncio fan-out in ``op_async`` for off-node workers.
 """
 from .collective import FederatedShardEngine, init_process_group_from_env, shard_slice  # noqa: F401
-from .streams import MultiShardDispatcher  # noqa: F401
+from .streams import MultiShardDispatcher, NativeMultiShardEngine  # noqa: F401
