@@ -31,7 +31,7 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=200)
     p.add_argument("--warmup", type=int, default=20)
-    p.add_argument("--model", choices=["linear", "logistic"], default="linear")
+    p.add_argument("--model", choices=["linear", "logistic", "ode"], default="linear")
     p.add_argument("--rows", type=int, default=None, help="rows per GPU shard")
     p.add_argument("--features", type=int, default=1024, help="logistic GLM features")
     p.add_argument("--dtype", choices=["bf16", "f32", "f64"], default="bf16")
@@ -91,6 +91,26 @@ def main():
         theta0 = np.array([1.5, 0.5])
         config_model = "gaussian_linear_regression"
         shape_cfg = {"rows_per_gpu": rows}
+    elif args.model == "ode":
+        from pytensor_federated_amd.models import ODEModel
+        from pytensor_federated_amd.models.ode import (
+            generate_ode_dataset,
+            lotka_volterra_rhs,
+        )
+
+        n_exp = 1024 if have_gpu else 16
+        n_steps = 50
+        u0, obs_idx, y_obs = generate_ode_dataset(
+            n_experiments=n_exp, n_obs=20, n_steps=n_steps, t1=8.0, sigma=0.1,
+            seed=1234 + rank,
+        )
+        model = ODEModel(
+            lotka_volterra_rhs, u0, 0.0, 8.0, n_steps, obs_idx, y_obs, 0.1,
+            device=device, dtype=torch.float64,
+        )
+        theta0 = np.array([0.8, 0.3, 0.6, 0.2])
+        config_model = "lotka_volterra_ode_adjoint"
+        shape_cfg = {"experiments_per_gpu": n_exp, "rk4_steps": n_steps, "n_obs": 20}
     else:
         k = args.features
         X = torch.randn(rows, k, generator=gen) / math.sqrt(k)
@@ -179,7 +199,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": args.dtype if have_gpu else "f32",
+            "dtype": "f64" if args.model == "ode" else (args.dtype if have_gpu else "f32"),
             "data": "synthetic",
             "config": {
                 "model": config_model,
