@@ -129,9 +129,22 @@ def main():
     theta_dev = torch.as_tensor(theta0, dtype=torch.float64, device=device)
     host_buf = None
 
+    # graphed linear path: hipGraph replay of [H2D theta -> fused kernel ->
+    # RCCL all-reduce -> mailbox publish] per evaluation
+    graphed = None
+    if args.model == "linear" and have_gpu and use_kernels is None and readback:
+        try:
+            from pytensor_federated_amd.parallel.graphed import GraphedLinearEngine
+
+            graphed = GraphedLinearEngine(model, distributed=distributed)
+        except Exception as ex:
+            print(f"# graphed path unavailable ({ex}); falling back", flush=True)
+            graphed = None
+
     # single-GPU linear serving path: one native sync call per evaluation
     fast_sync = (
-        not distributed
+        graphed is None
+        and not distributed
         and readback
         and args.model == "linear"
         and hasattr(model, "logp_grad_sync")
@@ -142,6 +155,8 @@ def main():
         # every rank derives the same perturbed theta (the broadcast of theta
         # from the driver is folded into the all-reduce round trip below)
         scale = 1.0 + 0.001 * math.sin(t)
+        if graphed is not None:
+            return graphed.logp_grad_sync(a0 * scale, b0 * scale)
         if fast_sync:
             return model.logp_grad_sync(a0 * scale, b0 * scale)
         if args.model == "linear":
@@ -209,6 +224,11 @@ def main():
                 "readback": readback,
                 "device": str(device),
                 "kernels": bool(use_kernels is None),
+                "path": (
+                    "hipgraph-replay" if graphed is not None
+                    else "sync-native" if fast_sync
+                    else "engine"
+                ),
                 **shape_cfg,
             },
         }

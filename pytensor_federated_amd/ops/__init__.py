@@ -70,6 +70,18 @@ def _try_load() -> Optional[ctypes.CDLL]:
     ]
     lib.fed_host_alloc.restype = ctypes.c_void_p
     lib.fed_host_alloc.argtypes = [ctypes.c_longlong]
+    lib.fed_gaussian_linear_theta.restype = ctypes.c_int
+    lib.fed_gaussian_linear_theta.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_longlong,
+        ctypes.c_void_p, ctypes.c_double,
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_longlong,
+        ctypes.c_int, ctypes.c_void_p,
+    ]
+    lib.fed_publish_result.restype = ctypes.c_int
+    lib.fed_publish_result.argtypes = [
+        ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.c_void_p,
+    ]
     lib.fed_logistic_glm.restype = ctypes.c_int
     lib.fed_logistic_glm.argtypes = [
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_longlong, ctypes.c_int,
@@ -170,6 +182,44 @@ def _get_mailbox() -> "_np.ndarray":
         )
         _mailbox[:] = 0.0
     return _mailbox
+
+
+def alloc_mailbox(n_results: int) -> "_np.ndarray":
+    """A fresh mapped-pinned mailbox: n_results fp64 + one u64 seq slot."""
+    lib = require_kernels()
+    ptr = lib.fed_host_alloc((n_results + 1) * 8)
+    if not ptr:
+        raise RuntimeError("hipHostMalloc failed")
+    arr = _np.ctypeslib.as_array(
+        ctypes.cast(ptr, ctypes.POINTER(ctypes.c_double)), shape=(n_results + 1,)
+    )
+    arr[:] = 0.0
+    return arr
+
+
+def gaussian_linear_launch_theta(
+    x: torch.Tensor, y: torch.Tensor, theta_dev: torch.Tensor, sigma: float,
+    out: torch.Tensor, ws: torch.Tensor,
+) -> None:
+    """Async fused launch reading [a,b] from device memory (graph-capturable)."""
+    lib = require_kernels()
+    rc = lib.fed_gaussian_linear_theta(
+        x.data_ptr(), y.data_ptr(), x.numel(),
+        theta_dev.data_ptr(), float(sigma),
+        out.data_ptr(), ws.data_ptr(), ws.numel() * 8,
+        _DTYPE_CODE[x.dtype], _stream_ptr(),
+    )
+    _check(rc, "fed_gaussian_linear_theta")
+
+
+def publish_result(buf: torch.Tensor, mailbox: "_np.ndarray", epoch_dev: torch.Tensor) -> None:
+    """Launch the mailbox-publish kernel (graph-capturable; epoch on device)."""
+    lib = require_kernels()
+    rc = lib.fed_publish_result(
+        buf.data_ptr(), buf.numel(), mailbox.ctypes.data,
+        epoch_dev.data_ptr(), _stream_ptr(),
+    )
+    _check(rc, "fed_publish_result")
 
 
 def gaussian_workspace(device) -> torch.Tensor:

@@ -164,13 +164,15 @@ __global__ __launch_bounds__(256) void k_gaussian_linear_fused(
     unsigned* __restrict__ ticket,   // monotonic arrival counter
     double* __restrict__ out3,       // device result
     double* __restrict__ out3_host,  // mapped pinned mailbox (nullable)
-    unsigned long long seq           // call sequence for the mailbox flag
+    unsigned long long seq,          // call sequence for the mailbox flag
+    const double* __restrict__ theta_dev  // optional [a, b] device buffer
 ) {
     using TR = VecTraits<T>;
     using A = typename TR::acc_t;
     constexpr int VEC = TR::VEC;
-    const double a = a_d;
-    const double b = b_d;
+    // theta from device memory enables hipGraph replay with varying theta
+    const double a = theta_dev ? theta_dev[0] : a_d;
+    const double b = theta_dev ? theta_dev[1] : b_d;
 
     const long long gid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
     const long long gstride = (long long)gridDim.x * blockDim.x;
@@ -454,7 +456,8 @@ static int gaussian_linear_impl(
     double a, double b, double sigma,
     double* out3, double* out3_host,
     double* workspace, long long ws_bytes,
-    int dtype, hipStream_t stream, unsigned long long seq
+    int dtype, hipStream_t stream, unsigned long long seq,
+    const double* theta_dev = nullptr
 ) {
     const double inv_sig2 = 1.0 / (sigma * sigma);
     const double logp_const = -0.5 * (double)n * log(2.0 * M_PI * sigma * sigma);
@@ -477,17 +480,17 @@ static int gaussian_linear_impl(
         case FED_F32:
             hipLaunchKernelGGL(k_gaussian_linear_fused<float>, dim3(grid), dim3(block), 0, stream,
                                (const float*)x, (const float*)y, n, a, b, inv_sig2, logp_const,
-                               slab, ticket, out3, out3_host, seq);
+                               slab, ticket, out3, out3_host, seq, theta_dev);
             break;
         case FED_F64:
             hipLaunchKernelGGL(k_gaussian_linear_fused<double>, dim3(grid), dim3(block), 0, stream,
                                (const double*)x, (const double*)y, n, a, b, inv_sig2, logp_const,
-                               slab, ticket, out3, out3_host, seq);
+                               slab, ticket, out3, out3_host, seq, theta_dev);
             break;
         case FED_BF16:
             hipLaunchKernelGGL(k_gaussian_linear_fused<bf16_tag>, dim3(grid), dim3(block), 0, stream,
                                (const bf16_tag*)x, (const bf16_tag*)y, n, a, b, inv_sig2, logp_const,
-                               slab, ticket, out3, out3_host, seq);
+                               slab, ticket, out3, out3_host, seq, theta_dev);
             break;
     }
     return (int)hipGetLastError();
@@ -599,6 +602,36 @@ int fed_logistic_glm(
 
 }  // extern "C"
 
+
+// Publish a small fp64 result vector to the pinned mailbox with a
+// self-incrementing epoch flag -- replayable from a hipGraph (the epoch
+// lives in device memory, not in baked kernel args).
+__global__ __launch_bounds__(64) void k_publish_result(
+    const double* __restrict__ buf, int n,
+    double* __restrict__ mailbox,            // [n results][1 seq slot]
+    unsigned long long* __restrict__ epoch
+) {
+    if (threadIdx.x == 0) {
+        for (int k = 0; k < n; ++k) mailbox[k] = buf[k];
+        const unsigned long long e = *epoch + 1;
+        *epoch = e;
+        __threadfence_system();
+        ((unsigned long long*)mailbox)[n] = e;
+    }
+}
+
+extern "C" int fed_publish_result(
+    const double* buf, int n, double* mailbox_host,
+    unsigned long long* epoch_dev, void* stream_v
+) {
+    void* mailbox_dev = nullptr;
+    hipError_t perr = hipHostGetDevicePointer(&mailbox_dev, mailbox_host, 0);
+    if (perr != hipSuccess) return (int)perr;
+    hipLaunchKernelGGL(k_publish_result, dim3(1), dim3(64), 0,
+                       (hipStream_t)stream_v, buf, n, (double*)mailbox_dev, epoch_dev);
+    return (int)hipGetLastError();
+}
+
 // ---------------------------------------------------------------------------
 // Native multi-shard linear engine: N shards on one GPU, one call per eval
 // ---------------------------------------------------------------------------
@@ -652,6 +685,19 @@ struct FedLinearEngine {
 extern "C" {
 
 int fed_linear_engine_destroy(void* handle);
+
+// hipGraph-capturable launch: theta read from a device buffer at kernel
+// execution time (so a captured graph replays with updated theta).
+int fed_gaussian_linear_theta(
+    const void* x, const void* y, long long n,
+    const double* theta_dev, double sigma,
+    double* out3, double* workspace, long long ws_bytes,
+    int dtype, void* stream_v
+) {
+    return gaussian_linear_impl(x, y, n, 0.0, 0.0, sigma, out3, nullptr,
+                                workspace, ws_bytes, dtype, (hipStream_t)stream_v,
+                                0, theta_dev);
+}
 
 void* fed_linear_engine_create(
     int n_shards, const void** xs, const void** ys, const long long* ns,

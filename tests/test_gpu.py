@@ -194,3 +194,45 @@ def test_native_multi_shard_engine(dev):
             assert engine.logp_grad_sync(1.0, 0.5) == ref
     finally:
         engine.close()
+
+
+def test_graphed_linear_engine(dev):
+    from pytensor_federated_amd.parallel.graphed import GraphedLinearEngine
+
+    x, y = generate_linear_dataset(2_000_000, seed=34)
+    m = GaussianLinearModel(x, y, sigma=0.4, device=dev, dtype=torch.bfloat16, use_kernels=True)
+    eng = GraphedLinearEngine(m, distributed=False)
+    for a, b in [(1.0, 0.5), (0.3, -0.2), (2.0, 1.0)]:
+        logp_g, ga_g, gb_g = eng.logp_grad_sync(a, b)
+        logp_r, ga_r, gb_r = m.logp_grad_sync(a, b)
+        np.testing.assert_allclose(logp_g, logp_r, rtol=1e-12)
+        np.testing.assert_allclose(ga_g, ga_r, rtol=1e-12)
+        np.testing.assert_allclose(gb_g, gb_r, rtol=1e-12)
+    # replay stability
+    ref = eng.logp_grad_sync(1.0, 0.5)
+    for _ in range(100):
+        assert eng.logp_grad_sync(1.0, 0.5) == ref
+
+
+def test_graphed_engine_with_nccl_world1(dev):
+    """Capture including the RCCL all_reduce (world_size 1 on this box;
+    the 8-GPU version is the same graph on every rank)."""
+    import torch.distributed as dist
+
+    if dist.is_initialized():
+        pytest.skip("process group already initialized")
+    dist.init_process_group(
+        "nccl", init_method="tcp://127.0.0.1:29537", rank=0, world_size=1
+    )
+    try:
+        from pytensor_federated_amd.parallel.graphed import GraphedLinearEngine
+
+        x, y = generate_linear_dataset(500_000, seed=35)
+        m = GaussianLinearModel(x, y, sigma=0.4, device=dev, dtype=torch.bfloat16)
+        eng = GraphedLinearEngine(m, distributed=True)
+        logp_g, ga_g, gb_g = eng.logp_grad_sync(1.2, 0.4)
+        logp_r, ga_r, gb_r = m.logp_grad_sync(1.2, 0.4)
+        np.testing.assert_allclose(logp_g, logp_r, rtol=1e-12)
+        np.testing.assert_allclose(ga_g, ga_r, rtol=1e-12)
+    finally:
+        dist.destroy_process_group()
