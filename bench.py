@@ -129,7 +129,7 @@ def main():
     theta_dev = torch.as_tensor(theta0, dtype=torch.float64, device=device)
     host_buf = None
 
-    # graphed linear path: hipGraph replay of [H2D theta -> fused kernel ->
+    # graphed paths: hipGraph replay of [H2D theta -> model eval ->
     # RCCL all-reduce -> mailbox publish] per evaluation
     graphed = None
     if args.model == "linear" and have_gpu and use_kernels is None and readback:
@@ -139,6 +139,14 @@ def main():
             graphed = GraphedLinearEngine(model, distributed=distributed)
         except Exception as ex:
             print(f"# graphed path unavailable ({ex}); falling back", flush=True)
+            graphed = None
+    elif args.model == "ode" and have_gpu and readback:
+        try:
+            from pytensor_federated_amd.parallel.graphed import GraphedLogpGradEngine
+
+            graphed = GraphedLogpGradEngine(model, (4,), distributed=distributed)
+        except Exception as ex:
+            print(f"# graphed ODE path unavailable ({ex}); falling back", flush=True)
             graphed = None
 
     # single-GPU linear serving path: one native sync call per evaluation
@@ -156,7 +164,9 @@ def main():
         # from the driver is folded into the all-reduce round trip below)
         scale = 1.0 + 0.001 * math.sin(t)
         if graphed is not None:
-            return graphed.logp_grad_sync(a0 * scale, b0 * scale)
+            if args.model == "linear":
+                return graphed.logp_grad_sync(a0 * scale, b0 * scale)
+            return graphed.logp_grad_sync(theta0 * scale)
         if fast_sync:
             return model.logp_grad_sync(a0 * scale, b0 * scale)
         if args.model == "linear":

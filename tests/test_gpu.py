@@ -236,3 +236,18 @@ def test_graphed_engine_with_nccl_world1(dev):
         np.testing.assert_allclose(ga_g, ga_r, rtol=1e-12)
     finally:
         dist.destroy_process_group()
+
+
+def test_graphed_ode_engine(dev):
+    from pytensor_federated_amd.models import ODEModel
+    from pytensor_federated_amd.models.ode import generate_ode_dataset, lotka_volterra_rhs
+    from pytensor_federated_amd.parallel.graphed import GraphedLogpGradEngine
+
+    u0, obs_idx, y_obs = generate_ode_dataset(n_experiments=32, n_obs=10, n_steps=30, t1=5.0)
+    m = ODEModel(lotka_volterra_rhs, u0, 0.0, 5.0, 30, obs_idx, y_obs, 0.1, device=dev)
+    eng = GraphedLogpGradEngine(m, (4,), distributed=False)
+    for theta in [np.array([0.8, 0.3, 0.6, 0.2]), np.array([0.7, 0.25, 0.65, 0.22])]:
+        logp_g, (grad_g,) = eng(theta)
+        logp_r, (grad_r,) = m(theta)
+        np.testing.assert_allclose(float(logp_g), float(logp_r), rtol=1e-10)
+        np.testing.assert_allclose(grad_g, grad_r, rtol=1e-8)
