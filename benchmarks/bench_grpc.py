@@ -27,7 +27,10 @@ def _serve(port: int, rows: int):
 
     x, y = generate_linear_dataset(rows, seed=0)
     model = GaussianLinearModel(x, y, sigma=0.4)
-    serve_compute_func(wrap_logp_grad_func(model.as_logp_grad_func()), "127.0.0.1", port)
+    serve_compute_func(
+        wrap_logp_grad_func(model.as_logp_grad_func()), "127.0.0.1", port,
+        fast_port=port + 1,
+    )
 
 
 def main():
@@ -36,6 +39,7 @@ def main():
     parser.add_argument("--warmup", type=int, default=100)
     parser.add_argument("--rows", type=int, default=10)
     parser.add_argument("--port", type=int, default=9651)
+    parser.add_argument("--transport", choices=["grpc", "fast"], default="grpc")
     parser.add_argument("--unary", action="store_true")
     args = parser.parse_args()
 
@@ -53,7 +57,10 @@ def main():
     try:
         from pytensor_federated_amd.common import LogpGradServiceClient
 
-        client = LogpGradServiceClient("127.0.0.1", args.port)
+        if args.transport == "fast":
+            client = LogpGradServiceClient("127.0.0.1", args.port + 1, transport="fast")
+        else:
+            client = LogpGradServiceClient("127.0.0.1", args.port)
         use_stream = not args.unary
         for _ in range(args.warmup):
             client.evaluate(1.5, 0.5, use_stream=use_stream)
@@ -68,7 +75,7 @@ def main():
                     "value": args.calls / elapsed,
                     "unit": "calls/s",
                     "ms_per_call": elapsed / args.calls * 1000,
-                    "transport": "unary" if args.unary else "bidirectional stream",
+                    "transport": args.transport + (" unary" if args.unary else " stream"),
                     "rows": args.rows,
                     "config": "BASELINE config 1",
                 }

@@ -1,0 +1,147 @@
+"""Fast-transport tests: same behaviors as the gRPC edge, lower latency."""
+import asyncio
+import multiprocessing
+import socket
+import time
+
+import numpy as np
+import pytest
+
+import pytensor_federated_amd.service as service_mod
+from pytensor_federated_amd.service import (
+    ArraysToArraysService,
+    ArraysToArraysServiceClient,
+    _privates,
+    get_load_async,
+    thread_pid_id,
+)
+
+FAST_PORTS = (9561, 9562)
+
+
+def _serve_fast(port: int):
+    import asyncio
+
+    from pytensor_federated_amd.fastsock import start_fast_server_async
+    from pytensor_federated_amd.service import ArraysToArraysService
+
+    async def main():
+        service = ArraysToArraysService(lambda a, b: [a * b])
+        server = await start_fast_server_async(service, "127.0.0.1", port)
+        async with server:
+            await server.serve_forever()
+
+    asyncio.run(main())
+
+
+def _wait_tcp(port, timeout=30.0):
+    deadline = time.time() + timeout
+    while time.time() < deadline:
+        try:
+            with socket.create_connection(("127.0.0.1", port), timeout=1):
+                return
+        except OSError:
+            time.sleep(0.1)
+    raise TimeoutError(f"port {port} never opened")
+
+
+@pytest.fixture(scope="module")
+def fast_servers():
+    ctx = multiprocessing.get_context("spawn")
+    procs = {p: ctx.Process(target=_serve_fast, args=(p,), daemon=True) for p in FAST_PORTS}
+    for proc in procs.values():
+        proc.start()
+    try:
+        for port in FAST_PORTS:
+            _wait_tcp(port)
+        yield procs
+    finally:
+        for proc in procs.values():
+            if proc.is_alive():
+                proc.terminate()
+        for proc in procs.values():
+            proc.join(timeout=10)
+
+
+@pytest.mark.timeout(120)
+def test_fast_evaluate_roundtrip(fast_servers):
+    client = ArraysToArraysServiceClient("127.0.0.1", FAST_PORTS[0], transport="fast")
+    a, b = np.array([1.0, 4.0]), np.array([2.0, 3.0])
+    for _ in range(5):
+        (out,) = client.evaluate(a, b)
+    np.testing.assert_array_equal(out, a * b)
+    del client
+
+
+@pytest.mark.timeout(120)
+def test_fast_get_load_and_n_clients(fast_servers):
+    load = asyncio.run(get_load_async("127.0.0.1", FAST_PORTS[0], transport="fast"))
+    assert load is not None
+    client = ArraysToArraysServiceClient("127.0.0.1", FAST_PORTS[0], transport="fast")
+    client.evaluate(np.array(2.0), np.array(3.0))  # opens persistent connection
+    load = asyncio.run(get_load_async("127.0.0.1", FAST_PORTS[0], transport="fast"))
+    assert load.n_clients >= 1
+    del client
+
+
+@pytest.mark.timeout(120)
+def test_fast_remote_error_propagates(fast_servers):
+    client = ArraysToArraysServiceClient("127.0.0.1", FAST_PORTS[0], transport="fast", retries=0)
+    with pytest.raises(RuntimeError, match="remote evaluation failed"):
+        client.evaluate(np.array(2.0))  # arity error on the worker
+    del client
+
+
+FAILOVER_FAST_PORTS = (9563, 9564)
+
+
+@pytest.mark.timeout(180)
+def test_fast_failover(monkeypatch):
+    # own servers: this test kills one of them
+    monkeypatch.setattr(service_mod, "_BALANCE_DESYNC_RANGE", (0.0, 0.01))
+    ctx = multiprocessing.get_context("spawn")
+    procs = {
+        p: ctx.Process(target=_serve_fast, args=(p,), daemon=True) for p in FAILOVER_FAST_PORTS
+    }
+    for proc in procs.values():
+        proc.start()
+    try:
+        for port in FAILOVER_FAST_PORTS:
+            _wait_tcp(port)
+        hap = [("127.0.0.1", p) for p in FAILOVER_FAST_PORTS]
+        client = ArraysToArraysServiceClient(hosts_and_ports=hap, transport="fast", retries=2)
+        a, b = np.array(2.0), np.array(5.0)
+        (out,) = client.evaluate(a, b)
+        np.testing.assert_array_equal(out, np.array(10.0))
+        connected_port = _privates[thread_pid_id(client)].port
+        procs[connected_port].terminate()
+        procs[connected_port].join()
+        (out,) = client.evaluate(a, b)
+        np.testing.assert_array_equal(out, np.array(10.0))
+        assert _privates[thread_pid_id(client)].port != connected_port
+        del client
+    finally:
+        for proc in procs.values():
+            if proc.is_alive():
+                proc.terminate()
+        for proc in procs.values():
+            proc.join(timeout=10)
+
+
+@pytest.mark.timeout(120)
+def test_fast_latency_beats_grpc_floor(fast_servers):
+    """The point of the transport: well under gRPC's ~1-2 ms container floor."""
+    client = ArraysToArraysServiceClient("127.0.0.1", FAST_PORTS[1], transport="fast")
+
+    async def run():
+        a, b = np.array(2.0), np.array(3.0)
+        for _ in range(20):
+            await client.evaluate_async(a, b)
+        t0 = time.perf_counter()
+        for _ in range(200):
+            await client.evaluate_async(a, b)
+        return (time.perf_counter() - t0) / 200
+
+    per_call = asyncio.run(run())
+    assert per_call < 0.002, f"fast transport too slow: {per_call * 1e6:.0f} us/call"
+    del client
