@@ -31,23 +31,45 @@ def needs_rebuild() -> bool:
     return any(src.stat().st_mtime > lib_mtime for src in sources())
 
 
+WORKER_SRC = CSRC / "fed_worker.cpp"
+WORKER_BIN = OPS_DIR / "fed_worker"
+
+
 def build(force: bool = False, verbose: bool = True) -> Path:
-    """Compile every .hip source into one shared library for gfx950."""
-    if not force and not needs_rebuild():
-        return LIB_PATH
-    cmd = [
-        HIPCC,
-        f"--offload-arch={ARCH}",
-        "-O3",
-        "-std=c++17",
-        "-shared",
-        "-fPIC",
-        "-o",
-        str(LIB_PATH),
-    ] + [str(s) for s in sources()]
-    if verbose:
-        print("[pytensor_federated_amd.ops.build]", " ".join(cmd))
-    subprocess.run(cmd, check=True)
+    """Compile the .hip sources into one shared library (gfx950) and the
+    native worker daemon binary."""
+    if force or needs_rebuild():
+        cmd = [
+            HIPCC,
+            f"--offload-arch={ARCH}",
+            "-O3",
+            "-std=c++17",
+            "-shared",
+            "-fPIC",
+            "-o",
+            str(LIB_PATH),
+        ] + [str(s) for s in sources()]
+        if verbose:
+            print("[pytensor_federated_amd.ops.build]", " ".join(cmd))
+        subprocess.run(cmd, check=True)
+    if WORKER_SRC.exists() and (
+        force
+        or not WORKER_BIN.exists()
+        or WORKER_SRC.stat().st_mtime > WORKER_BIN.stat().st_mtime
+    ):
+        cmd = [
+            HIPCC,
+            f"--offload-arch={ARCH}",
+            "-O3",
+            "-std=c++17",
+            str(WORKER_SRC),
+            "-ldl",
+            "-o",
+            str(WORKER_BIN),
+        ]
+        if verbose:
+            print("[pytensor_federated_amd.ops.build]", " ".join(cmd))
+        subprocess.run(cmd, check=True)
     return LIB_PATH
 
 
