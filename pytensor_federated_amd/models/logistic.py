@@ -126,6 +126,40 @@ class LogisticGLMModel(LogpGradModel):
             return out[0], [out[1:]]
         return logp, grads
 
+    def logp_grad_batched(self, theta) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Evaluate 16 chains at once: theta[K,16] -> (logp[16], G[K,16]).
+
+        On an MI355X shard this runs the MFMA-batched kernel (one pass over
+        X for all 16 proposal vectors); elsewhere an eager batched matmul.
+        """
+        theta = torch.as_tensor(theta)
+        if theta.dim() != 2 or theta.shape[0] != self._k:
+            raise ValueError(f"theta must be [{self._k}, B], got {tuple(theta.shape)}")
+        if (
+            self._kernel_path()
+            and self._X.dtype == torch.bfloat16
+            and theta.shape[1] == 16
+            and self._k in (512, 1024)
+        ):
+            from ..ops import logistic_glm_logp_grad_batched
+
+            return logistic_glm_logp_grad_batched(self._X, self._y, theta)
+        return self._logp_grad_batched_eager(theta)
+
+    def _logp_grad_batched_eager(self, theta: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+        X, y = self._X, self._y
+        acc_dtype = torch.float64 if X.dtype == torch.float64 else torch.float32
+        theta = theta.to(device=X.device, dtype=acc_dtype)
+        Xf = X.to(acc_dtype)
+        yf = y.to(acc_dtype)
+        Z = Xf @ theta
+        logp = torch.sum(
+            yf[:, None] * Z - torch.nn.functional.softplus(Z), dim=0, dtype=torch.float64
+        )
+        R = yf[:, None] - torch.sigmoid(Z)
+        G = (Xf.t() @ R).to(torch.float64)
+        return logp, G
+
     def _logp_grad_eager(self, beta: torch.Tensor) -> Tuple[torch.Tensor, List[torch.Tensor]]:
         X, y = self._X, self._y
         acc_dtype = torch.float64 if X.dtype == torch.float64 else torch.float32

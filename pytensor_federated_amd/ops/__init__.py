@@ -88,6 +88,12 @@ def _try_load() -> Optional[ctypes.CDLL]:
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_longlong,
         ctypes.c_int, ctypes.c_void_p,
     ]
+    lib.fed_logistic_glm_batched.restype = ctypes.c_int
+    lib.fed_logistic_glm_batched.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_longlong, ctypes.c_int,
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_longlong,
+        ctypes.c_void_p,
+    ]
     lib.fed_last_hip_error.restype = ctypes.c_char_p
     _lib = lib
     return _lib
@@ -220,6 +226,44 @@ def publish_result(buf: torch.Tensor, mailbox: "_np.ndarray", epoch_dev: torch.T
         epoch_dev.data_ptr(), _stream_ptr(),
     )
     _check(rc, "fed_publish_result")
+
+
+BATCH_CHAINS = 16
+
+
+def logistic_glm_logp_grad_batched(
+    X: torch.Tensor,
+    y: torch.Tensor,
+    theta: torch.Tensor,
+    out: Optional[torch.Tensor] = None,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """MFMA-batched evaluation of 16 chains: theta[K,16] -> logp[16], G[K,16].
+
+    One pass over X computes Z = X.theta, the per-chain BCE logp, and
+    G = X^T (y - sigmoid(Z)) on the matrix cores (kernel
+    k_logistic_glm_batched).  Per-chain cost ~B x below the single-chain
+    kernel -- the multi-chain MCMC axis (reference pm.sample cores=N) on
+    one GPU.
+    """
+    lib = require_kernels()
+    n, K = X.shape
+    B = BATCH_CHAINS
+    if theta.shape != (K, B):
+        raise ValueError(f"theta must be [K, {B}], got {tuple(theta.shape)}")
+    if X.dtype != torch.bfloat16:
+        raise TypeError("batched kernel supports bf16 X only")
+    assert X.is_cuda and X.is_contiguous()
+    theta_t = theta.detach().t().contiguous().to(device=X.device, dtype=torch.bfloat16)
+    if out is None:
+        out = torch.empty(B + K * B, dtype=torch.float64, device=X.device)
+    ws = _workspace(X.device, f"logistic_batched{K}", 304 * (B + K * B), dtype=torch.float32)
+    rc = lib.fed_logistic_glm_batched(
+        X.data_ptr(), y.data_ptr(), n, K,
+        theta_t.data_ptr(), out.data_ptr(), ws.data_ptr(), ws.numel() * 4,
+        _stream_ptr(),
+    )
+    _check(rc, "fed_logistic_glm_batched")
+    return out[:B], out[B:].reshape(K, B)
 
 
 def gaussian_workspace(device) -> torch.Tensor:

@@ -844,3 +844,295 @@ int fed_ipc_open(const unsigned char* handle64, void** dev_ptr) {
 int fed_ipc_close(void* dev_ptr) { return (int)hipIpcCloseMemHandle(dev_ptr); }
 
 }  // extern "C"
+
+// ---------------------------------------------------------------------------
+// MFMA fragment-map probe (test-only): D = A[16x32] . B[32x16], bf16->f32.
+// Assumed lane mapping (verified on hardware by tests/test_gpu.py):
+//   A: row i = lane&15, k = (lane>>4)*8 + j   (8 bf16 per lane)
+//   B: col j = lane&15, k = (lane>>4)*8 + jj  (8 bf16 per lane)
+//   D: col   = lane&15, row = (lane>>4)*4 + reg
+// ---------------------------------------------------------------------------
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+__global__ __launch_bounds__(64) void k_mfma_probe(
+    const unsigned short* __restrict__ A,  // [16][32] bf16 row-major
+    const unsigned short* __restrict__ B,  // [32][16] bf16 row-major
+    float* __restrict__ D                  // [16][16] f32 row-major
+) {
+    const int lane = threadIdx.x & 63;
+    union { bf16x8 v; unsigned short u[8]; } a_frag, b_frag;
+    const int arow = lane & 15;
+    const int k0 = (lane >> 4) * 8;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+        a_frag.u[j] = A[arow * 32 + k0 + j];
+        b_frag.u[j] = B[(k0 + j) * 16 + (lane & 15)];
+    }
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag.v, b_frag.v, acc, 0, 0, 0);
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        const int row = (lane >> 4) * 4 + r;
+        const int col = lane & 15;
+        D[row * 16 + col] = acc[r];
+    }
+}
+
+extern "C" int fed_mfma_probe(const void* A, const void* B, void* D, void* stream_v) {
+    hipLaunchKernelGGL(k_mfma_probe, dim3(1), dim3(64), 0, (hipStream_t)stream_v,
+                       (const unsigned short*)A, (const unsigned short*)B, (float*)D);
+    return (int)hipGetLastError();
+}
+
+// ---------------------------------------------------------------------------
+// Batched logistic GLM: B=16 chains per call on MFMA matrix cores
+// ---------------------------------------------------------------------------
+//
+// Multi-chain MCMC (the reference's pm.sample(cores=N) axis) evaluates 16
+// proposal vectors in ONE pass over X:
+//
+//     Z[rows,16]  = X[rows,K] . Theta[K,16]        (phase A, MFMA)
+//     logp[b]    += sum_r  y z - softplus(z)
+//     R[rows,16]  = y - sigmoid(Z)
+//     G[K,16]    += X^T[K,rows] . R[rows,16]       (phase B, MFMA)
+//
+// X is HBM-read once per call (k-chunks staged to LDS, phase B re-reads the
+// L2-hot chunk); at B=16 the arithmetic is 4*N*K*B flops -- VALU could not
+// keep up with the HBM stream (0.8 TFLOP/call vs ~4 ms of traffic), MFMA
+// makes compute a ~10% bystander.  Per-chain cost is ~B x lower than the
+// single-chain kernel.
+//
+// Geometry: block = 256 threads = 4 waves; row tile 64 (16 rows/wave in
+// phase A); K chunked by 128 (each wave owns a 32-col slice in phase B);
+// G accumulates in AGPRs (16 tiles x 4 f32/lane/wave); per-block partials
+// go to an fp32 slab [16 logp | K*16 G], reduced by k_colsum_reduce.
+//
+// Verified fragment maps (fed_mfma_probe): A[i=l&15][k=(l>>4)*8+j],
+// B[k=(l>>4)*8+j][j=l&15], D[col=l&15][row=(l>>4)*4+r].
+
+#define BCH 16           // chains per call
+#define BL_ROWS 64       // rows per block tile
+#define BL_CHUNK 128     // K columns per staged chunk
+#define XPAD 8           // X_lds row pad (elems): stride 272 B, b128-clean
+#define TPAD 8           // Theta row pad: stride 2064+16 B
+#define RPAD 8           // R_T row pad: stride 144 B
+
+template <int K>  // compile-time K: keeps g_acc statically indexed (no
+                  // scratch spill -- cdna_hip_programming.md rule 20)
+__global__ __launch_bounds__(256) void k_logistic_glm_batched(
+    const unsigned short* __restrict__ X,   // [N][K] bf16
+    const unsigned short* __restrict__ y,   // [N] bf16
+    long long n_rows,
+    const unsigned short* __restrict__ theta_t,  // [BCH][K] bf16 (transposed!)
+    float* __restrict__ slab                     // [grid][BCH + K*BCH]
+) {
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    constexpr int n_chunks = K / BL_CHUNK;
+
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    // one dynamic region, 16B-aligned carves (G17)
+    unsigned short* th_lds = (unsigned short*)smem;                 // [BCH][K+TPAD]
+    const int th_stride = K + TPAD;
+    unsigned short* x_lds = th_lds + BCH * th_stride;               // [BL_ROWS][BL_CHUNK+XPAD]
+    const int x_stride = BL_CHUNK + XPAD;
+    unsigned short* rt_lds = x_lds + BL_ROWS * x_stride;            // [BCH][BL_ROWS+RPAD]
+    const int rt_stride = BL_ROWS + RPAD;
+    float* y_lds = (float*)(rt_lds + BCH * rt_stride + 8);          // [BL_ROWS] (+8 aligns to 4B*?; carve kept 16B-ish)
+    float* red_lds = y_lds + BL_ROWS;                               // [256] scratch for logp reduce
+
+    // ---- stage Theta^T once per block (straight copy, b128 rows) ----
+    for (int idx = threadIdx.x * 8; idx < BCH * K; idx += 256 * 8) {
+        const int b = idx / K;
+        const int k = idx % K;
+        *(U4*)&th_lds[b * th_stride + k] = *(const U4*)&theta_t[b * K + k];
+    }
+    __syncthreads();
+
+    typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
+    typedef __attribute__((ext_vector_type(4))) float f32x4_t;
+    union frag_u { bf16x8_t v; unsigned short u[8]; U4 q; };
+
+    // G accumulators: wave owns k-cols [c*128 + wid*32, +32) of each chunk
+    f32x4_t g_acc[n_chunks * 2];
+#pragma unroll
+    for (int t = 0; t < n_chunks * 2; ++t) g_acc[t] = (f32x4_t){0.f, 0.f, 0.f, 0.f};
+    float logp_acc = 0.f;  // this lane's chain partial (chain = lane&15)
+
+    const long long n_tiles = (n_rows + BL_ROWS - 1) / BL_ROWS;
+    for (long long tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
+        const long long row0 = tile * BL_ROWS;
+        // ---- stage y tile ----
+        if (threadIdx.x < BL_ROWS) {
+            const long long r = row0 + threadIdx.x;
+            y_lds[threadIdx.x] = r < n_rows ? bf16_bits_to_f32(y[r]) : 0.f;
+        }
+
+        // ---- phase A: Z = X . Theta (accumulate over chunks) ----
+        f32x4_t z_acc = {0.f, 0.f, 0.f, 0.f};  // wave's 16 rows x 16 chains
+#pragma unroll
+        for (int c = 0; c < n_chunks; ++c) {
+            // stage chunk [BL_ROWS][BL_CHUNK]: thread loads 16B; 16 thr/row
+            __syncthreads();
+            {
+                const int r = threadIdx.x / 16;          // 0..15
+                const int kk = (threadIdx.x % 16) * 8;   // 0..120
+#pragma unroll
+                for (int rr = 0; rr < 4; ++rr) {
+                    const long long row = row0 + r + rr * 16;
+                    U4 val = {0, 0, 0, 0};
+                    if (row < n_rows)
+                        val = *(const U4*)&X[row * (long long)K + c * BL_CHUNK + kk];
+                    *(U4*)&x_lds[(r + rr * 16) * x_stride + kk] = val;
+                }
+            }
+            __syncthreads();
+            // wave computes its 16 rows (rows wid*16..+16) over this chunk
+#pragma unroll
+            for (int ks = 0; ks < BL_CHUNK / 32; ++ks) {
+                frag_u a, b;
+                const int arow = wid * 16 + (lane & 15);
+                const int ak = c * 0 + ks * 32 + (lane >> 4) * 8;  // within chunk
+                a.q = *(U4*)&x_lds[arow * x_stride + ak];
+                const int bk = c * BL_CHUNK + ks * 32 + (lane >> 4) * 8;  // absolute k
+                b.q = *(U4*)&th_lds[(lane & 15) * th_stride + bk];
+                // note: B fragment wants B[k][j]: our th_lds is [chain][k] --
+                // reading 8 contiguous k at fixed chain gives B^T fragment,
+                // which IS B[k][j] with j = chain = lane&15. Contraction dim
+                // k comes from the register index -- matches the map.
+                z_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, z_acc, 0, 0, 0);
+            }
+        }
+
+        // ---- logp + R from Z ----
+        __syncthreads();  // x_lds free; write R_T
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int row_in_wave = (lane >> 4) * 4 + r;      // 0..15
+            const int row_in_tile = wid * 16 + row_in_wave;   // 0..63
+            const long long row = row0 + row_in_tile;
+            const int chain = lane & 15;
+            float z = z_acc[r];
+            float yv = y_lds[row_in_tile];
+            float resid = 0.f;
+            if (row < n_rows) {
+                const float sp = fmaxf(z, 0.f) + log1pf(__expf(-fabsf(z)));
+                logp_acc += yv * z - sp;
+                resid = yv - 1.f / (1.f + __expf(-z));
+            }
+            // R^T[chain][row]
+            unsigned short rb;
+            union { float f; unsigned int u; } cv;
+            cv.f = resid;
+            const unsigned int rnd = 0x7fff + ((cv.u >> 16) & 1);
+            rb = (unsigned short)((cv.u + rnd) >> 16);
+            rt_lds[chain * rt_stride + row_in_tile] = rb;
+        }
+
+        // ---- phase B: G += X_chunk^T . R (re-stage chunks; L2-hot) ----
+#pragma unroll
+        for (int c = 0; c < n_chunks; ++c) {
+            __syncthreads();
+            {
+                const int r = threadIdx.x / 16;
+                const int kk = (threadIdx.x % 16) * 8;
+#pragma unroll
+                for (int rr = 0; rr < 4; ++rr) {
+                    const long long row = row0 + r + rr * 16;
+                    U4 val = {0, 0, 0, 0};
+                    if (row < n_rows)
+                        val = *(const U4*)&X[row * (long long)K + c * BL_CHUNK + kk];
+                    *(U4*)&x_lds[(r + rr * 16) * x_stride + kk] = val;
+                }
+            }
+            __syncthreads();
+            // wave's 32 k-cols of this chunk: 2 MFMA tiles x 2 k-steps
+#pragma unroll
+            for (int t2 = 0; t2 < 2; ++t2) {
+                const int kcol0 = wid * 32 + t2 * 16;  // within chunk
+                f32x4_t acc = g_acc[c * 2 + t2];
+#pragma unroll
+                for (int rs = 0; rs < 2; ++rs) {  // contraction: rows 0..31, 32..63
+                    frag_u a, b;
+                    // A = X^T fragment: [kcol][row]; 8 rows strided in LDS
+                    const int kcol = kcol0 + (lane & 15);
+                    const int arow0 = rs * 32 + (lane >> 4) * 8;
+#pragma unroll
+                    for (int j = 0; j < 8; ++j)
+                        a.u[j] = x_lds[(arow0 + j) * x_stride + kcol];
+                    // B = R fragment: [row][chain] from R^T[chain][row]
+                    b.q = *(U4*)&rt_lds[(lane & 15) * rt_stride + rs * 32 + (lane >> 4) * 8];
+                    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, acc, 0, 0, 0);
+                }
+                g_acc[c * 2 + t2] = acc;
+            }
+        }
+    }
+
+    // ---- epilogue: block partials -> slab ----
+    // logp: lane partial for chain lane&15; reduce across (lane>>4) and waves
+    red_lds[threadIdx.x] = logp_acc;
+    __syncthreads();
+    float* slab_blk = slab + (long long)blockIdx.x * (BCH + (long long)K * BCH);
+    if (threadIdx.x < BCH) {
+        float s = 0.f;
+        for (int i = threadIdx.x; i < 256; i += BCH) s += red_lds[i];
+        slab_blk[threadIdx.x] = s;
+    }
+    // G: lane holds chains=lane&15, kcols=(lane>>4)*4+r for each tile
+    float* g_slab = slab_blk + BCH;
+#pragma unroll
+    for (int c = 0; c < n_chunks; ++c) {
+#pragma unroll
+        for (int t2 = 0; t2 < 2; ++t2) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int kcol = c * BL_CHUNK + wid * 32 + t2 * 16 + (lane >> 4) * 4 + r;
+                const int chain = lane & 15;
+                g_slab[(long long)kcol * BCH + chain] = g_acc[c * 2 + t2][r];
+            }
+        }
+    }
+}
+
+extern "C" int fed_logistic_glm_batched(
+    const void* X, const void* y, long long n_rows, int K,
+    const void* theta_t_bf16,  // [16][K] bf16, transposed theta
+    double* out,               // fp64[16 + K*16]: [logp[16] | G[K][16]]
+    float* workspace, long long ws_bytes,
+    void* stream_v
+) {
+    hipStream_t stream = (hipStream_t)stream_v;
+    if (K != 512 && K != 1024) return -4;
+    const int block = 256;
+    int grid = pick_grid(n_rows / BL_ROWS + 1, 1);
+    if (grid > 304) grid = 304;  // slab size cap; >256 CUs covered
+    const long long slab_cols = BCH + (long long)K * BCH;
+    if ((long long)grid * slab_cols * 4 > ws_bytes)
+        grid = (int)(ws_bytes / (slab_cols * 4));
+    if (grid < 1) return -3;
+    const int lds_bytes =
+        (BCH * (K + TPAD) + BL_ROWS * (BL_CHUNK + XPAD) + BCH * (BL_ROWS + RPAD) + 8) * 2 +
+        (BL_ROWS + 256) * 4 + 64;
+    if (K == 1024)
+        hipLaunchKernelGGL(k_logistic_glm_batched<1024>, dim3(grid), dim3(block), lds_bytes,
+                           stream, (const unsigned short*)X, (const unsigned short*)y, n_rows,
+                           (const unsigned short*)theta_t_bf16, workspace);
+    else
+        hipLaunchKernelGGL(k_logistic_glm_batched<512>, dim3(grid), dim3(block), lds_bytes,
+                           stream, (const unsigned short*)X, (const unsigned short*)y, n_rows,
+                           (const unsigned short*)theta_t_bf16, workspace);
+    hipError_t kerr = hipGetLastError();
+    if (kerr != hipSuccess) return (int)kerr;
+    const int rgrid = ((int)slab_cols + 255) / 256;
+    int chunks = grid / 8;
+    if (chunks < 1) chunks = 1;
+    if (chunks > 64) chunks = 64;
+    hipError_t merr = hipMemsetAsync(out, 0, slab_cols * 8, stream);
+    if (merr != hipSuccess) return (int)merr;
+    hipLaunchKernelGGL(k_colsum_reduce, dim3(rgrid, chunks), dim3(256), 0, stream,
+                       workspace, grid, (int)slab_cols, out);
+    return (int)hipGetLastError();
+}

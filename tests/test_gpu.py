@@ -251,3 +251,55 @@ def test_graphed_ode_engine(dev):
         logp_r, (grad_r,) = m(theta)
         np.testing.assert_allclose(float(logp_g), float(logp_r), rtol=1e-10)
         np.testing.assert_allclose(grad_g, grad_r, rtol=1e-8)
+
+
+class TestBatchedLogisticMFMA:
+    @pytest.mark.parametrize("K", [512, 1024])
+    def test_matches_eager_batched(self, dev, K):
+        X, y, _ = generate_logistic_dataset(20_000, K, seed=61)
+        m = LogisticGLMModel(X, y, device=dev, dtype=torch.bfloat16, use_kernels=True)
+        rng = np.random.RandomState(62)
+        theta = torch.as_tensor(rng.standard_normal((K, 16)) * 0.3).to(torch.bfloat16)
+        # eager reference on the SAME bf16-quantized theta/data
+        eager = LogisticGLMModel(m._X, m._y, device=dev, dtype=torch.bfloat16, use_kernels=False)
+        logp_e, G_e = eager._logp_grad_batched_eager(theta.float())
+        logp_k, G_k = m.logp_grad_batched(theta.float())
+        np.testing.assert_allclose(
+            logp_k.cpu().numpy(), logp_e.cpu().numpy(), rtol=5e-3
+        )
+        # G tolerance: the kernel quantizes residuals to bf16 for the MFMA
+        ge = G_e.cpu().numpy()
+        gk = G_k.cpu().numpy()
+        scale = np.abs(ge).max()
+        np.testing.assert_allclose(gk, ge, atol=2e-2 * scale, rtol=2e-2)
+
+    def test_batched_consistent_with_single_chain(self, dev):
+        K = 512
+        X, y, beta0 = generate_logistic_dataset(8_192, K, seed=63)
+        m = LogisticGLMModel(X, y, device=dev, dtype=torch.bfloat16, use_kernels=True)
+        theta = torch.as_tensor(np.tile(beta0[:, None], (1, 16))).float()
+        # quantize like the batched kernel does before comparing
+        theta_q = theta.to(torch.bfloat16).float()
+        logp_b, G_b = m.logp_grad_batched(theta_q)
+        logp_s, (g_s,) = m.logp_grad(theta_q[:, 0].contiguous())
+        for b in range(16):
+            np.testing.assert_allclose(float(logp_b[b]), float(logp_s), rtol=5e-3)
+        scale = float(torch.abs(g_s).max())
+        np.testing.assert_allclose(
+            G_b[:, 0].cpu().numpy(), g_s.cpu().numpy(), atol=2e-2 * scale, rtol=2e-2
+        )
+
+    def test_row_tail_handling(self, dev):
+        # N not a multiple of 64: pad rows must not contribute
+        K = 512
+        X, y, _ = generate_logistic_dataset(1000 + 17, K, seed=64)
+        m = LogisticGLMModel(X, y, device=dev, dtype=torch.bfloat16, use_kernels=True)
+        theta = torch.randn(K, 16).to(torch.bfloat16).float() * 0.2
+        eager = LogisticGLMModel(m._X, m._y, device=dev, dtype=torch.bfloat16, use_kernels=False)
+        logp_e, G_e = eager._logp_grad_batched_eager(theta)
+        logp_k, G_k = m.logp_grad_batched(theta)
+        np.testing.assert_allclose(logp_k.cpu().numpy(), logp_e.cpu().numpy(), rtol=5e-3)
+        scale = float(torch.abs(G_e).max())
+        np.testing.assert_allclose(
+            G_k.cpu().numpy(), G_e.cpu().numpy(), atol=2e-2 * scale, rtol=3e-2
+        )
