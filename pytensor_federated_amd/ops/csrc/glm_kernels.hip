@@ -933,85 +933,81 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched(
     constexpr int n_chunks = K / BL_CHUNK;
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    // one dynamic region, 16B-aligned carves (G17)
     unsigned short* th_lds = (unsigned short*)smem;                 // [BCH][K+TPAD]
     const int th_stride = K + TPAD;
-    unsigned short* x_lds = th_lds + BCH * th_stride;               // [BL_ROWS][BL_CHUNK+XPAD]
+    unsigned short* x_lds = th_lds + BCH * th_stride;               // [2][BL_ROWS][BL_CHUNK+XPAD]
     const int x_stride = BL_CHUNK + XPAD;
-    unsigned short* rt_lds = x_lds + BL_ROWS * x_stride;            // [BCH][BL_ROWS+RPAD]
+    const int x_buf = BL_ROWS * x_stride;
+    unsigned short* rt_lds = x_lds + 2 * x_buf;                     // [BCH][BL_ROWS+RPAD]
     const int rt_stride = BL_ROWS + RPAD;
-    float* y_lds = (float*)(rt_lds + BCH * rt_stride + 8);          // [BL_ROWS] (+8 aligns to 4B*?; carve kept 16B-ish)
-    float* red_lds = y_lds + BL_ROWS;                               // [256] scratch for logp reduce
+    float* y_lds = (float*)(rt_lds + BCH * rt_stride + 8);          // [BL_ROWS]
+    float* red_lds = y_lds + BL_ROWS;                               // [256]
 
-    // ---- stage Theta^T once per block (straight copy, b128 rows) ----
+    // ---- stage Theta^T once per block ----
     for (int idx = threadIdx.x * 8; idx < BCH * K; idx += 256 * 8) {
         const int b = idx / K;
         const int k = idx % K;
         *(U4*)&th_lds[b * th_stride + k] = *(const U4*)&theta_t[b * K + k];
     }
-    __syncthreads();
 
     typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
     typedef __attribute__((ext_vector_type(4))) float f32x4_t;
     union frag_u { bf16x8_t v; unsigned short u[8]; U4 q; };
 
-    // G accumulators: wave owns k-cols [c*128 + wid*32, +32) of each chunk
     f32x4_t g_acc[n_chunks * 2];
 #pragma unroll
     for (int t = 0; t < n_chunks * 2; ++t) g_acc[t] = (f32x4_t){0.f, 0.f, 0.f, 0.f};
     float logp_acc = 0.f;  // this lane's chain partial (chain = lane&15)
 
+    // staging helpers: thread owns rows {r0, r0+16, r0+32, r0+48}, 16 B each
+    const int st_r0 = threadIdx.x / 16;          // 0..15
+    const int st_k0 = (threadIdx.x % 16) * (BL_CHUNK / 16);
+
     const long long n_tiles = (n_rows + BL_ROWS - 1) / BL_ROWS;
     for (long long tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
         const long long row0 = tile * BL_ROWS;
-        // ---- stage y tile ----
         if (threadIdx.x < BL_ROWS) {
             const long long r = row0 + threadIdx.x;
             y_lds[threadIdx.x] = r < n_rows ? bf16_bits_to_f32(y[r]) : 0.f;
         }
 
-        // ---- phase A: Z = X . Theta (accumulate over chunks) ----
-        f32x4_t z_acc = {0.f, 0.f, 0.f, 0.f};  // wave's 16 rows x 16 chains
+        // double-buffered chunk pipeline over BOTH phases: phase A consumes
+        // chunks 0..n-1 (Z), phase B consumes them again (G).  One barrier
+        // per step; next chunk's global loads issue before the MFMAs.
+        U4 ld[4];
+#define LOAD_CHUNK(c)                                                              _Pragma("unroll") for (int rr = 0; rr < 4; ++rr) {                                 const long long row = row0 + st_r0 + rr * 16;                                  ld[rr] = (U4){0, 0, 0, 0};                                                     if (row < n_rows)                                                                  ld[rr] = *(const U4*)&X[row * (long long)K + (c) * BL_CHUNK + st_k0];     }
+#define WRITE_CHUNK(buf)                                                           _Pragma("unroll") for (int rr = 0; rr < 4; ++rr)                                   *(U4*)&x_lds[(buf) * x_buf + (st_r0 + rr * 16) * x_stride + st_k0] = ld[rr];
+
+        LOAD_CHUNK(0)
+        WRITE_CHUNK(0)
+        int cur = 0;
+
+        // ---- phase A: Z = X . Theta ----
+        f32x4_t z_acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int c = 0; c < n_chunks; ++c) {
-            // stage chunk [BL_ROWS][BL_CHUNK]: thread loads 16B; 16 thr/row
-            __syncthreads();
-            {
-                const int r = threadIdx.x / 16;          // 0..15
-                const int kk = (threadIdx.x % 16) * 8;   // 0..120
-#pragma unroll
-                for (int rr = 0; rr < 4; ++rr) {
-                    const long long row = row0 + r + rr * 16;
-                    U4 val = {0, 0, 0, 0};
-                    if (row < n_rows)
-                        val = *(const U4*)&X[row * (long long)K + c * BL_CHUNK + kk];
-                    *(U4*)&x_lds[(r + rr * 16) * x_stride + kk] = val;
-                }
-            }
-            __syncthreads();
-            // wave computes its 16 rows (rows wid*16..+16) over this chunk
+            __syncthreads();  // buf[cur] (and Theta on c==0) visible
+            if (c + 1 < n_chunks) LOAD_CHUNK(c + 1)
 #pragma unroll
             for (int ks = 0; ks < BL_CHUNK / 32; ++ks) {
                 frag_u a, b;
                 const int arow = wid * 16 + (lane & 15);
-                const int ak = c * 0 + ks * 32 + (lane >> 4) * 8;  // within chunk
-                a.q = *(U4*)&x_lds[arow * x_stride + ak];
-                const int bk = c * BL_CHUNK + ks * 32 + (lane >> 4) * 8;  // absolute k
+                const int ak = ks * 32 + (lane >> 4) * 8;
+                a.q = *(U4*)&x_lds[cur * x_buf + arow * x_stride + ak];
+                const int bk = c * BL_CHUNK + ks * 32 + (lane >> 4) * 8;
                 b.q = *(U4*)&th_lds[(lane & 15) * th_stride + bk];
-                // note: B fragment wants B[k][j]: our th_lds is [chain][k] --
-                // reading 8 contiguous k at fixed chain gives B^T fragment,
-                // which IS B[k][j] with j = chain = lane&15. Contraction dim
-                // k comes from the register index -- matches the map.
                 z_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, z_acc, 0, 0, 0);
             }
+            if (c + 1 < n_chunks) WRITE_CHUNK(cur ^ 1)
+            cur ^= 1;
         }
 
-        // ---- logp + R from Z ----
-        __syncthreads();  // x_lds free; write R_T
+        // ---- logp + R from Z (also prefetch chunk 0 for phase B) ----
+        LOAD_CHUNK(0)
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
-            const int row_in_wave = (lane >> 4) * 4 + r;      // 0..15
-            const int row_in_tile = wid * 16 + row_in_wave;   // 0..63
+            const int row_in_wave = (lane >> 4) * 4 + r;
+            const int row_in_tile = wid * 16 + row_in_wave;
             const long long row = row0 + row_in_tile;
             const int chain = lane & 15;
             float z = z_acc[r];
@@ -1022,57 +1018,46 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched(
                 logp_acc += yv * z - sp;
                 resid = yv - 1.f / (1.f + __expf(-z));
             }
-            // R^T[chain][row]
-            unsigned short rb;
             union { float f; unsigned int u; } cv;
             cv.f = resid;
             const unsigned int rnd = 0x7fff + ((cv.u >> 16) & 1);
-            rb = (unsigned short)((cv.u + rnd) >> 16);
-            rt_lds[chain * rt_stride + row_in_tile] = rb;
+            rt_lds[chain * rt_stride + row_in_tile] = (unsigned short)((cv.u + rnd) >> 16);
         }
+        __syncthreads();  // R complete; x_lds free
+        WRITE_CHUNK(0)
+        cur = 0;
 
-        // ---- phase B: G += X_chunk^T . R (re-stage chunks; L2-hot) ----
+        // ---- phase B: G += X_chunk^T . R (chunks are L2-hot) ----
 #pragma unroll
         for (int c = 0; c < n_chunks; ++c) {
             __syncthreads();
-            {
-                const int r = threadIdx.x / 16;
-                const int kk = (threadIdx.x % 16) * 8;
-#pragma unroll
-                for (int rr = 0; rr < 4; ++rr) {
-                    const long long row = row0 + r + rr * 16;
-                    U4 val = {0, 0, 0, 0};
-                    if (row < n_rows)
-                        val = *(const U4*)&X[row * (long long)K + c * BL_CHUNK + kk];
-                    *(U4*)&x_lds[(r + rr * 16) * x_stride + kk] = val;
-                }
-            }
-            __syncthreads();
-            // wave's 32 k-cols of this chunk: 2 MFMA tiles x 2 k-steps
+            if (c + 1 < n_chunks) LOAD_CHUNK(c + 1)
 #pragma unroll
             for (int t2 = 0; t2 < 2; ++t2) {
-                const int kcol0 = wid * 32 + t2 * 16;  // within chunk
+                const int kcol0 = wid * 32 + t2 * 16;
                 f32x4_t acc = g_acc[c * 2 + t2];
 #pragma unroll
-                for (int rs = 0; rs < 2; ++rs) {  // contraction: rows 0..31, 32..63
+                for (int rs = 0; rs < 2; ++rs) {
                     frag_u a, b;
-                    // A = X^T fragment: [kcol][row]; 8 rows strided in LDS
                     const int kcol = kcol0 + (lane & 15);
                     const int arow0 = rs * 32 + (lane >> 4) * 8;
 #pragma unroll
                     for (int j = 0; j < 8; ++j)
-                        a.u[j] = x_lds[(arow0 + j) * x_stride + kcol];
-                    // B = R fragment: [row][chain] from R^T[chain][row]
+                        a.u[j] = x_lds[cur * x_buf + (arow0 + j) * x_stride + kcol];
                     b.q = *(U4*)&rt_lds[(lane & 15) * rt_stride + rs * 32 + (lane >> 4) * 8];
                     acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, acc, 0, 0, 0);
                 }
                 g_acc[c * 2 + t2] = acc;
             }
+            if (c + 1 < n_chunks) WRITE_CHUNK(cur ^ 1)
+            cur ^= 1;
         }
+        __syncthreads();  // rt_lds reuse next tile
+#undef LOAD_CHUNK
+#undef WRITE_CHUNK
     }
 
     // ---- epilogue: block partials -> slab ----
-    // logp: lane partial for chain lane&15; reduce across (lane>>4) and waves
     red_lds[threadIdx.x] = logp_acc;
     __syncthreads();
     float* slab_blk = slab + (long long)blockIdx.x * (BCH + (long long)K * BCH);
@@ -1081,7 +1066,6 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched(
         for (int i = threadIdx.x; i < 256; i += BCH) s += red_lds[i];
         slab_blk[threadIdx.x] = s;
     }
-    // G: lane holds chains=lane&15, kcols=(lane>>4)*4+r for each tile
     float* g_slab = slab_blk + BCH;
 #pragma unroll
     for (int c = 0; c < n_chunks; ++c) {
@@ -1114,7 +1098,7 @@ extern "C" int fed_logistic_glm_batched(
         grid = (int)(ws_bytes / (slab_cols * 4));
     if (grid < 1) return -3;
     const int lds_bytes =
-        (BCH * (K + TPAD) + BL_ROWS * (BL_CHUNK + XPAD) + BCH * (BL_ROWS + RPAD) + 8) * 2 +
+        (BCH * (K + TPAD) + 2 * BL_ROWS * (BL_CHUNK + XPAD) + BCH * (BL_ROWS + RPAD) + 8) * 2 +
         (BL_ROWS + 256) * 4 + 64;
     if (K == 1024)
         hipLaunchKernelGGL(k_logistic_glm_batched<1024>, dim3(grid), dim3(block), lds_bytes,
