@@ -8,6 +8,7 @@ nothing beyond this package: every driver consumes the framework's
 ``LogpGradServiceClient.evaluate``, or a ``FederatedShardEngine`` all plug
 in unchanged (the logp+grad of one MCMC step is ONE fused worker call).
 """
+from .mala import sample_mala_batched  # noqa: F401
 from .map import find_map  # noqa: F401
 from .mcmc import Metropolis, sample_metropolis  # noqa: F401
 from .nuts import NUTS, sample_nuts  # noqa: F401

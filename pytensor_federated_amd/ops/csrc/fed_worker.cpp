@@ -32,7 +32,10 @@
 
 #include <hip/hip_runtime.h>
 
+#include <mutex>
+
 #include <string>
+#include <thread>
 #include <vector>
 
 // ---------------------------------------------------------------------------
@@ -280,6 +283,7 @@ static void serve_client(Worker& w, int fd) {
         memcpy(&ln, hdr + 1, 4);
         payload.resize(ln);
         if (ln && !read_exact(fd, payload.data(), ln)) break;
+        static std::mutex eval_mu;
         if (hdr[0] == 0x01) {  // Evaluate
             std::vector<ParsedArray> items;
             std::string uuid;
@@ -289,9 +293,15 @@ static void serve_client(Worker& w, int fd) {
             }
             const double a = scalar_value(items[0]);
             const double b = scalar_value(items[1]);
-            w.seq++;
-            int rc = w.eval(w.x_dev, w.y_dev, w.n, a, b, w.sigma, w.out_dev, w.mailbox,
+            int rc;
+            double res[3];
+            {
+                std::lock_guard<std::mutex> lock(eval_mu);
+                w.seq++;
+                rc = w.eval(w.x_dev, w.y_dev, w.n, a, b, w.sigma, w.out_dev, w.mailbox,
                             w.ws_dev, (72 + 3 * 2048) * 8, w.dtype, nullptr, w.seq);
+                res[0] = w.mailbox[0]; res[1] = w.mailbox[1]; res[2] = w.mailbox[2];
+            }
             if (rc != 0) {
                 char msg[64];
                 snprintf(msg, sizeof(msg), "kernel eval failed (%d)", rc);
@@ -299,9 +309,9 @@ static void serve_client(Worker& w, int fd) {
                 continue;
             }
             std::string out;
-            put_len_field(out, 1, encode_f64_scalar(w.mailbox[0]));  // logp
-            put_len_field(out, 1, encode_f64_scalar(w.mailbox[1]));  // d/da
-            put_len_field(out, 1, encode_f64_scalar(w.mailbox[2]));  // d/db
+            put_len_field(out, 1, encode_f64_scalar(res[0]));  // logp
+            put_len_field(out, 1, encode_f64_scalar(res[1]));  // d/da
+            put_len_field(out, 1, encode_f64_scalar(res[2]));  // d/db
             put_len_field(out, 2, uuid);
             if (!write_frame(fd, 0x81, out)) break;
         } else if (hdr[0] == 0x02) {  // GetLoad
@@ -380,7 +390,8 @@ int main(int argc, char** argv) {
     while (true) {
         int fd = accept(srv, nullptr, nullptr);
         if (fd < 0) continue;
-        serve_client(w, fd);  // one client at a time (federated workers are 1:1)
+        // one thread per client; evaluations serialize on the eval mutex
+        std::thread([&w, fd] { serve_client(w, fd); }).detach();
     }
     return 0;
 }
