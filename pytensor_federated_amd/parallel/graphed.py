@@ -16,7 +16,7 @@ same replay.
 from __future__ import annotations
 
 import time
-from typing import Optional, Tuple
+from typing import Tuple
 
 import numpy as np
 import torch

@@ -14,7 +14,7 @@ test suite for numerical equivalence).
 """
 from __future__ import annotations
 
-from typing import List, Optional, Sequence, Tuple
+from typing import List, Sequence, Tuple
 
 import torch
 

@@ -30,7 +30,7 @@ from typing import Dict, List, Optional, Sequence, Tuple
 
 import numpy as np
 
-from . import rpc
+
 from .npproto.utils import ndarray_from_numpy, ndarray_to_numpy
 from .rpc import (
     GetLoadParams,

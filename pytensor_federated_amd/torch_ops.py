@@ -14,7 +14,7 @@ PyTensor adapters with the reference's exact Op API live in
 """
 from __future__ import annotations
 
-from typing import Callable, List, Sequence
+from typing import Callable, Sequence
 
 import numpy as np
 import torch

@@ -286,8 +286,10 @@ def register_fuse_asyncs() -> None:
     """
     from pytensor.compile.mode import optdb
 
-    if "fuse_asyncs" not in optdb:
+    try:
         optdb.register("fuse_asyncs", AsyncFusionOptimizer(), "fast_run", position=90)
+    except Exception:
+        pass  # already registered (module re-import)
 
 
 register_fuse_asyncs()
