@@ -64,6 +64,7 @@ class ODEModel(LogpGradModel):
         device=None,
         dtype: torch.dtype = torch.float64,
         delay: Optional[float] = None,
+        use_kernels: Optional[bool] = None,
     ) -> None:
         """
         Parameters
@@ -100,6 +101,45 @@ class ODEModel(LogpGradModel):
         self._obs_components = obs_components
         self._sigma = float(sigma)
         self._dtype = dtype
+        self._use_kernels = use_kernels
+        self._native_state = None  # lazy (obs_of_step, states_ws, out)
+
+    def _native_path(self) -> bool:
+        """Native CDNA4 forward+adjoint kernels (Lotka-Volterra family only):
+        one lane integrates one experiment's whole trajectory in-register."""
+        want = self._u0.is_cuda if self._use_kernels is None else self._use_kernels
+        return (
+            want
+            and self._u0.is_cuda
+            and self.f is lotka_volterra_rhs
+            and self._obs_components is None
+            and self._dtype == torch.float64
+            and self._u0.shape[-1] == 2
+        )
+
+    def _logp_grad_native(self, theta: torch.Tensor):
+        import math as _math
+
+        from ..ops import ode_lv_logp_grad
+
+        if self._native_state is None:
+            obs_of_step = torch.full((self._n_steps + 1,), -1, dtype=torch.int32)
+            for j, idx in enumerate(self._obs_idx):
+                obs_of_step[idx] = j
+            B = self._u0.shape[0]
+            self._native_state = (
+                obs_of_step.to(self._u0.device),
+                torch.empty((self._n_steps + 1) * B * 2, dtype=torch.float64, device=self._u0.device),
+                torch.empty(5, dtype=torch.float64, device=self._u0.device),
+            )
+        obs_of_step, states_ws, out = self._native_state
+        ode_lv_logp_grad(
+            self._u0, self._y, obs_of_step, self._n_steps, self._h, self._sigma,
+            theta, states_ws, out=out,
+        )
+        n_vals = self._y.numel()
+        logp_const = -0.5 * n_vals * _math.log(2.0 * _math.pi * self._sigma**2)
+        return out[0] + logp_const, [out[1:5]]
 
     @property
     def device(self):
@@ -122,6 +162,8 @@ class ODEModel(LogpGradModel):
 
     def logp_grad(self, theta) -> Tuple[torch.Tensor, List[torch.Tensor]]:
         theta = torch.as_tensor(theta).to(device=self.device, dtype=self._dtype)
+        if self._native_path():
+            return self._logp_grad_native(theta)
         states = self._forward_states(theta)
 
         sig2 = self._sigma * self._sigma

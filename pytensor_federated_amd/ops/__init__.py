@@ -88,6 +88,12 @@ def _try_load() -> Optional[ctypes.CDLL]:
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_longlong,
         ctypes.c_int, ctypes.c_void_p,
     ]
+    lib.fed_ode_lv_eval.restype = ctypes.c_int
+    lib.fed_ode_lv_eval.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_int, ctypes.c_int, ctypes.c_double, ctypes.c_double,
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+    ]
     lib.fed_logistic_glm_batched.restype = ctypes.c_int
     lib.fed_logistic_glm_batched.argtypes = [
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_longlong, ctypes.c_int,
@@ -226,6 +232,36 @@ def publish_result(buf: torch.Tensor, mailbox: "_np.ndarray", epoch_dev: torch.T
         epoch_dev.data_ptr(), _stream_ptr(),
     )
     _check(rc, "fed_publish_result")
+
+
+def ode_lv_logp_grad(
+    u0: torch.Tensor,          # [B, 2] f64
+    y_obs: torch.Tensor,       # [n_obs, B, 2] f64
+    obs_of_step: torch.Tensor, # [n_steps+1] int32 (obs index or -1)
+    n_steps: int,
+    h: float,
+    sigma: float,
+    theta: torch.Tensor,       # [4] f64 (any device)
+    states_ws: torch.Tensor,   # [(n_steps+1)*B*2] f64 scratch
+    out: Optional[torch.Tensor] = None,  # f64[5]
+) -> torch.Tensor:
+    """Native Lotka-Volterra forward+adjoint: out = [logp_quad, g_theta[4]].
+
+    (The logp normalization constant is added by the caller.)
+    """
+    lib = require_kernels()
+    B = u0.shape[0]
+    theta_dev = theta.detach().to(device=u0.device, dtype=torch.float64).contiguous()
+    if out is None:
+        out = torch.empty(5, dtype=torch.float64, device=u0.device)
+    rc = lib.fed_ode_lv_eval(
+        u0.data_ptr(), y_obs.data_ptr(), obs_of_step.data_ptr(),
+        int(n_steps), int(B), float(h), float(sigma),
+        theta_dev.data_ptr(), states_ws.data_ptr(), out.data_ptr(),
+        _stream_ptr(),
+    )
+    _check(rc, "fed_ode_lv_eval")
+    return out
 
 
 BATCH_CHAINS = 16

@@ -303,3 +303,38 @@ class TestBatchedLogisticMFMA:
         np.testing.assert_allclose(
             G_k.cpu().numpy(), G_e.cpu().numpy(), atol=2e-2 * scale, rtol=3e-2
         )
+
+
+class TestNativeODE:
+    def test_native_matches_generic_adjoint(self, dev):
+        from pytensor_federated_amd.models import ODEModel
+        from pytensor_federated_amd.models.ode import generate_ode_dataset, lotka_volterra_rhs
+
+        u0, obs_idx, y = generate_ode_dataset(n_experiments=64, n_obs=12, n_steps=40, t1=6.0)
+        theta0 = np.array([0.8, 0.3, 0.6, 0.2])
+        native = ODEModel(lotka_volterra_rhs, u0, 0.0, 6.0, 40, obs_idx, y, 0.1,
+                          device=dev, use_kernels=True)
+        generic = ODEModel(lotka_volterra_rhs, u0, 0.0, 6.0, 40, obs_idx, y, 0.1,
+                           device=dev, use_kernels=False)
+        assert native._native_path() and not generic._native_path()
+        logp_n, (g_n,) = native(theta0)
+        logp_g, (g_g,) = generic(theta0)
+        np.testing.assert_allclose(float(logp_n), float(logp_g), rtol=1e-12)
+        np.testing.assert_allclose(g_n, g_g, rtol=1e-9)
+
+    def test_native_ode_speed(self, dev):
+        import time
+
+        from pytensor_federated_amd.models import ODEModel
+        from pytensor_federated_amd.models.ode import generate_ode_dataset, lotka_volterra_rhs
+
+        u0, obs_idx, y = generate_ode_dataset(n_experiments=1024, n_obs=20, n_steps=50, t1=8.0)
+        theta0 = np.array([0.8, 0.3, 0.6, 0.2])
+        m = ODEModel(lotka_volterra_rhs, u0, 0.0, 8.0, 50, obs_idx, y, 0.1, device=dev)
+        m(theta0)
+        t0 = time.perf_counter()
+        for _ in range(50):
+            m(theta0)
+        per = (time.perf_counter() - t0) / 50
+        print(f"native ODE eval: {per * 1e6:.0f} us/call")
+        assert per < 0.005  # generic path was 92 ms eager / 25.6 ms graphed
