@@ -1178,7 +1178,9 @@ __global__ __launch_bounds__(256) void k_gaussian_persistent(
     const long long nvec = n / VEC;
     __shared__ double lds[4 * 3 + 4];  // reduce scratch + {seq, a, b} bcast
 
+#define PK_STAMP(code)     if (blockIdx.x == 0 && threadIdx.x == 0) ((volatile unsigned long long*)res_host)[4] = (code);
     unsigned long long my_seq = 0;
+    PK_STAMP(1)
     while (true) {
         // ---- acquire next request --------------------------------------
         if (threadIdx.x == 0) {
@@ -1233,6 +1235,7 @@ __global__ __launch_bounds__(256) void k_gaussian_persistent(
         }
         __syncthreads();
         if (lds[12] < 0.0) return;  // quit or spin give-up
+        PK_STAMP(2)
         const double a = lds[13];
         const double b = lds[14];
         __syncthreads();  // lds reused by the reduction below
@@ -1260,6 +1263,7 @@ __global__ __launch_bounds__(256) void k_gaussian_persistent(
             srx += r * xi;
             sr2 += r * r;
         }
+        PK_STAMP(3)
         double acc[3] = {sr2, sr, srx};
         block_reduce_add<3>(acc, lds);
         bool last = false;
@@ -1294,6 +1298,7 @@ __global__ __launch_bounds__(256) void k_gaussian_persistent(
             __syncthreads();
             block_reduce_add<3>(fin, lds);
             if (threadIdx.x == 0) {
+                ((volatile unsigned long long*)res_host)[5] = my_seq;
                 res_host[0] = logp_const - 0.5 * inv_sig2 * fin[0];
                 res_host[1] = inv_sig2 * fin[1];
                 res_host[2] = inv_sig2 * fin[2];
@@ -1302,6 +1307,7 @@ __global__ __launch_bounds__(256) void k_gaussian_persistent(
                 store_sc1_u64(&st->done_seq, my_seq);
             }
         }
+        PK_STAMP(4)
         // ---- completion barrier (slab must not be reused early) ---------
         if (threadIdx.x == 0) {
             long long spins = 0;
@@ -1337,12 +1343,12 @@ void* fed_gaussian_persistent_start(
         hipMemset(e->ws, 0, ws_words * 8) != hipSuccess ||
         hipMalloc(&e->st, sizeof(PersistentState)) != hipSuccess ||
         hipMemset(e->st, 0, sizeof(PersistentState)) != hipSuccess ||
-        hipHostMalloc((void**)&e->req, 4 * 8, hipHostMallocMapped) != hipSuccess ||
-        hipHostMalloc((void**)&e->res, 4 * 8, hipHostMallocMapped) != hipSuccess) {
+        hipHostMalloc((void**)&e->req, 8 * 8, hipHostMallocMapped) != hipSuccess ||
+        hipHostMalloc((void**)&e->res, 8 * 8, hipHostMallocMapped) != hipSuccess) {
         delete e;
         return nullptr;
     }
-    for (int i = 0; i < 4; ++i) {
+    for (int i = 0; i < 8; ++i) {
         e->req[i] = 0.0;
         e->res[i] = 0.0;
     }
@@ -1403,6 +1409,15 @@ int fed_gaussian_persistent_eval(void* handle, double a, double b, double* out3)
         }
     }
     return -6;  // server did not answer (it self-exits on its own spin bound)
+}
+
+int fed_gaussian_persistent_debug(void* handle, double* req8, double* res8) {
+    FedPersistentLinear* e = (FedPersistentLinear*)handle;
+    for (int i = 0; i < 8; ++i) {
+        req8[i] = e->req[i];
+        res8[i] = e->res[i];
+    }
+    return 0;
 }
 
 int fed_gaussian_persistent_stop(void* handle) {
