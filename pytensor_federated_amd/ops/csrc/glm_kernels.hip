@@ -1298,12 +1298,15 @@ __global__ __launch_bounds__(256) void k_gaussian_persistent(
             __syncthreads();
             block_reduce_add<3>(fin, lds);
             if (threadIdx.x == 0) {
-                ((volatile unsigned long long*)res_host)[5] = my_seq;
-                res_host[0] = logp_const - 0.5 * inv_sig2 * fin[0];
-                res_host[1] = inv_sig2 * fin[1];
-                res_host[2] = inv_sig2 * fin[2];
+                // volatile: inside the persistent loop the compiler defers /
+                // elides plain mailbox stores (measured: data landed, the
+                // plain u64 flag never did)
+                volatile double* rh = (volatile double*)res_host;
+                rh[0] = logp_const - 0.5 * inv_sig2 * fin[0];
+                rh[1] = inv_sig2 * fin[1];
+                rh[2] = inv_sig2 * fin[2];
                 __threadfence_system();
-                ((unsigned long long*)res_host)[3] = my_seq;
+                ((volatile unsigned long long*)res_host)[3] = my_seq;
                 store_sc1_u64(&st->done_seq, my_seq);
             }
         }
