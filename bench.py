@@ -129,10 +129,30 @@ def main():
     theta_dev = torch.as_tensor(theta0, dtype=torch.float64, device=device)
     host_buf = None
 
+    # persistent eval-server path (single-GPU linear): resident kernel,
+    # pinned request/result mailboxes -- no launch, no ramp per call
+    persistent = None
+    if (
+        args.model == "linear"
+        and have_gpu
+        and not distributed
+        and use_kernels is None
+        and readback
+    ):
+        try:
+            from pytensor_federated_amd.ops import PersistentLinearEngine
+
+            persistent = PersistentLinearEngine(model._x, model._y, model._sigma)
+        except Exception as ex:
+            print(f"# persistent path unavailable ({ex}); falling back", flush=True)
+            persistent = None
+
     # graphed paths: hipGraph replay of [H2D theta -> model eval ->
     # RCCL all-reduce -> mailbox publish] per evaluation
     graphed = None
-    if args.model == "linear" and have_gpu and use_kernels is None and readback:
+    if persistent is not None:
+        pass
+    elif args.model == "linear" and have_gpu and use_kernels is None and readback:
         try:
             from pytensor_federated_amd.parallel.graphed import GraphedLinearEngine
 
@@ -163,6 +183,8 @@ def main():
         # every rank derives the same perturbed theta (the broadcast of theta
         # from the driver is folded into the all-reduce round trip below)
         scale = 1.0 + 0.001 * math.sin(t)
+        if persistent is not None:
+            return persistent.logp_grad_sync(a0 * scale, b0 * scale)
         if graphed is not None:
             if args.model == "linear":
                 return graphed.logp_grad_sync(a0 * scale, b0 * scale)
@@ -235,7 +257,8 @@ def main():
                 "device": str(device),
                 "kernels": bool(use_kernels is None),
                 "path": (
-                    "hipgraph-replay" if graphed is not None
+                    "persistent-kernel" if persistent is not None
+                    else "hipgraph-replay" if graphed is not None
                     else "sync-native" if fast_sync
                     else "engine"
                 ),

@@ -1340,6 +1340,13 @@ void* fed_gaussian_persistent_start(
     const void* x, const void* y, long long n, double sigma, int dtype
 ) {
     FedPersistentLinear* e = new FedPersistentLinear();
+    {
+        const char* env = getenv("FED_PERSIST_GRID");
+        if (env) {
+            int g = atoi(env);
+            if (g >= 8 && g <= 1024 && (g & (g - 1)) == 0) e->grid = g;
+        }
+    }
     const long long ws_words = 72 + 3 * (long long)e->grid;
     if (hipStreamCreateWithFlags(&e->stream, hipStreamNonBlocking) != hipSuccess ||
         hipMalloc(&e->ws, ws_words * 8) != hipSuccess ||

@@ -338,3 +338,21 @@ class TestNativeODE:
         per = (time.perf_counter() - t0) / 50
         print(f"native ODE eval: {per * 1e6:.0f} us/call")
         assert per < 0.005  # generic path was 92 ms eager / 25.6 ms graphed
+
+
+def test_persistent_linear_engine(dev):
+    from pytensor_federated_amd.ops import PersistentLinearEngine
+
+    x, y = generate_linear_dataset(2_000_000, seed=93)
+    m = GaussianLinearModel(x, y, sigma=0.4, device=dev, dtype=torch.bfloat16, use_kernels=True)
+    eng = PersistentLinearEngine(m._x, m._y, 0.4)
+    try:
+        for a, b in [(1.5, 0.5), (0.3, -0.2)]:
+            got = eng.logp_grad_sync(a, b)
+            ref = m.logp_grad_sync(a, b)
+            np.testing.assert_allclose(got, ref, rtol=1e-12)
+        ref = eng.logp_grad_sync(1.0, 0.5)
+        for _ in range(300):
+            assert eng.logp_grad_sync(1.0, 0.5) == ref
+    finally:
+        eng.close()
