@@ -211,7 +211,9 @@ def main():
             import torch.distributed as dist
 
             dist.barrier()
-        if have_gpu:
+        if have_gpu and persistent is None:
+            # (the persistent eval server answers synchronously; a device-wide
+            # synchronize would instead wait out its idle lifetime)
             torch.cuda.synchronize()
 
     for t in range(args.warmup):

@@ -1141,7 +1141,12 @@ extern "C" int fed_logistic_glm_batched(
 // and shuts the server down.
 
 #define PK_SENTINEL 0xFFFFFFFFFFFFFFFFull
-#define PK_SPIN_LIMIT 20000000ll  // give-up bound: ~10-60 s of s_sleep polling
+// Spin bounds. The REQUEST poll doubles as the idle lifetime: a resident
+// kernel blocks any device-wide synchronize, so it exits after ~1-2 s idle
+// and the host relaunches it transparently (hipStreamQuery detects exit).
+#define PK_REQ_SPIN_LIMIT 500000ll    // ~1-2 s idle -> self-exit
+#define PK_SPIN_LIMIT 800000ll        // worker bcast poll (~1-2 s)
+#define PK_DONE_SPIN_LIMIT 200000ll   // completion barrier (compute is us-scale)
 
 struct PersistentState {       // device memory
     unsigned long long bcast_seq;   // sc1-published request broadcast
@@ -1196,7 +1201,7 @@ __global__ __launch_bounds__(256) void k_gaussian_persistent(
                     if (quit) { next = PK_SENTINEL; break; }
                     if (rs >= next) break;
                     __builtin_amdgcn_s_sleep(32);
-                    if (++spins > PK_SPIN_LIMIT) { next = PK_SENTINEL; break; }
+                    if (++spins > PK_REQ_SPIN_LIMIT) { next = PK_SENTINEL; break; }
                 }
                 if (next != PK_SENTINEL) {
                     // sc1 payload + drained sc1 flag (G16 R1: a plain store
@@ -1316,7 +1321,7 @@ __global__ __launch_bounds__(256) void k_gaussian_persistent(
             long long spins = 0;
             while (load_sc1_u64(&st->done_seq) < my_seq) {
                 __builtin_amdgcn_s_sleep(8);
-                if (++spins > PK_SPIN_LIMIT) { lds[12] = -1.0; break; }
+                if (++spins > PK_DONE_SPIN_LIMIT) { lds[12] = -1.0; break; }
             }
         }
         __syncthreads();
@@ -1332,7 +1337,44 @@ struct FedPersistentLinear {
     hipStream_t stream = nullptr;
     unsigned long long seq = 0;
     int grid = 512;
+    // launch parameters (for transparent relaunch after idle self-exit)
+    const void* x = nullptr;
+    const void* y = nullptr;
+    long long n = 0;
+    double inv_sig2 = 0;
+    double logp_const = 0;
+    int dtype = FED_BF16;
+    void* req_dev = nullptr;
+    void* res_dev = nullptr;
 };
+
+static int persistent_launch(FedPersistentLinear* e) {
+    unsigned* ticket = (unsigned*)e->ws;
+    double* slab = e->ws + 72;
+    switch (e->dtype) {
+        case FED_BF16:
+            hipLaunchKernelGGL(k_gaussian_persistent<bf16_tag>, dim3(e->grid), dim3(256), 0,
+                               e->stream, (const bf16_tag*)e->x, (const bf16_tag*)e->y, e->n,
+                               e->inv_sig2, e->logp_const, slab, ticket, e->st,
+                               (const volatile double*)e->req_dev, (double*)e->res_dev);
+            break;
+        case FED_F32:
+            hipLaunchKernelGGL(k_gaussian_persistent<float>, dim3(e->grid), dim3(256), 0,
+                               e->stream, (const float*)e->x, (const float*)e->y, e->n,
+                               e->inv_sig2, e->logp_const, slab, ticket, e->st,
+                               (const volatile double*)e->req_dev, (double*)e->res_dev);
+            break;
+        case FED_F64:
+            hipLaunchKernelGGL(k_gaussian_persistent<double>, dim3(e->grid), dim3(256), 0,
+                               e->stream, (const double*)e->x, (const double*)e->y, e->n,
+                               e->inv_sig2, e->logp_const, slab, ticket, e->st,
+                               (const volatile double*)e->req_dev, (double*)e->res_dev);
+            break;
+        default:
+            return -2;
+    }
+    return (int)hipGetLastError();
+}
 
 extern "C" {
 
@@ -1369,34 +1411,15 @@ void* fed_gaussian_persistent_start(
         delete e;
         return nullptr;
     }
-    const double inv_sig2 = 1.0 / (sigma * sigma);
-    const double logp_const = -0.5 * (double)n * log(2.0 * M_PI * sigma * sigma);
-    unsigned* ticket = (unsigned*)e->ws;
-    double* slab = e->ws + 72;
-    switch (dtype) {
-        case FED_BF16:
-            hipLaunchKernelGGL(k_gaussian_persistent<bf16_tag>, dim3(e->grid), dim3(256), 0,
-                               e->stream, (const bf16_tag*)x, (const bf16_tag*)y, n,
-                               inv_sig2, logp_const, slab, ticket, e->st,
-                               (const volatile double*)req_dev, (double*)res_dev);
-            break;
-        case FED_F32:
-            hipLaunchKernelGGL(k_gaussian_persistent<float>, dim3(e->grid), dim3(256), 0,
-                               e->stream, (const float*)x, (const float*)y, n,
-                               inv_sig2, logp_const, slab, ticket, e->st,
-                               (const volatile double*)req_dev, (double*)res_dev);
-            break;
-        case FED_F64:
-            hipLaunchKernelGGL(k_gaussian_persistent<double>, dim3(e->grid), dim3(256), 0,
-                               e->stream, (const double*)x, (const double*)y, n,
-                               inv_sig2, logp_const, slab, ticket, e->st,
-                               (const volatile double*)req_dev, (double*)res_dev);
-            break;
-        default:
-            delete e;
-            return nullptr;
-    }
-    if (hipGetLastError() != hipSuccess) {
+    e->x = x;
+    e->y = y;
+    e->n = n;
+    e->dtype = dtype;
+    e->inv_sig2 = 1.0 / (sigma * sigma);
+    e->logp_const = -0.5 * (double)n * log(2.0 * M_PI * sigma * sigma);
+    e->req_dev = req_dev;
+    e->res_dev = res_dev;
+    if (persistent_launch(e) != 0) {
         delete e;
         return nullptr;
     }
@@ -1410,15 +1433,24 @@ int fed_gaussian_persistent_eval(void* handle, double a, double b, double* out3)
     e->req[2] = b;
     __atomic_store_n((unsigned long long*)&e->req[0], e->seq, __ATOMIC_RELEASE);
     volatile unsigned long long* flag = ((volatile unsigned long long*)e->res) + 3;
-    for (long long spins = 0; spins < 2000000000LL; ++spins) {
+    int relaunches = 0;
+    for (long long spins = 0; spins < 4000000000LL; ++spins) {
         if (*flag >= e->seq) {
             out3[0] = e->res[0];
             out3[1] = e->res[1];
             out3[2] = e->res[2];
             return 0;
         }
+        if ((spins & 0xFFFFF) == 0xFFFFF) {  // every ~1M spins (~1-2 ms)
+            // server may have idle-exited (bounded request poll); relaunch --
+            // the fresh kernel sees the pending req_seq and serves it
+            if (hipStreamQuery(e->stream) != hipErrorNotReady) {
+                if (relaunches++ > 4) return -6;
+                if (persistent_launch(e) != 0) return -7;
+            }
+        }
     }
-    return -6;  // server did not answer (it self-exits on its own spin bound)
+    return -6;
 }
 
 int fed_gaussian_persistent_debug(void* handle, double* req8, double* res8) {
