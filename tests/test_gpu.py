@@ -356,3 +356,27 @@ def test_persistent_linear_engine(dev):
             assert eng.logp_grad_sync(1.0, 0.5) == ref
     finally:
         eng.close()
+
+
+def test_persistent_engine_relaunch_after_idle(dev):
+    """The resident server self-exits after ~1-2 s idle (so device-wide
+    synchronizes can't block on it); the next eval must transparently
+    relaunch it and still return the correct result."""
+    import time
+
+    from pytensor_federated_amd.ops import PersistentLinearEngine
+
+    x, y = generate_linear_dataset(500_000, seed=94)
+    m = GaussianLinearModel(x, y, sigma=0.4, device=dev, dtype=torch.bfloat16, use_kernels=True)
+    eng = PersistentLinearEngine(m._x, m._y, 0.4)
+    try:
+        ref = eng.logp_grad_sync(1.2, 0.4)
+        time.sleep(4.0)  # > idle lifetime: server exits
+        torch.cuda.synchronize()  # must not block now
+        got = eng.logp_grad_sync(1.2, 0.4)  # transparent relaunch
+        assert got == ref
+        # and the relaunched server keeps serving
+        for _ in range(50):
+            assert eng.logp_grad_sync(1.2, 0.4) == ref
+    finally:
+        eng.close()
