@@ -144,6 +144,71 @@ __device__ inline void block_reduce_add(double* vals, double* lds /* [4][NACC] *
 // Gaussian linear regression: fused logp + d/da + d/db
 // ---------------------------------------------------------------------------
 
+
+// Shared element loop for the gaussian reductions.  Mixed precision: the
+// residual r is computed in f64 (an f32 pred carries a systematic rounding
+// bias that pollutes the cancellation-heavy sum(r) -- measured ~1e-3
+// relative at N=1e7), but the r^2 / r*x PRODUCTS accumulate in f32: their
+// rounding is uncorrelated (CLT-cancelling), and per-lane spans are short
+// before the f64 cross-lane tree.  Saves ~1/3 of the per-element f64 VALU.
+template <typename T>
+__device__ __forceinline__ void gauss_accumulate(
+    const T* __restrict__ x, const T* __restrict__ y, long long n,
+    long long gid, long long gstride, double a, double b,
+    double& sr_out, double& srx_out, double& sr2_out
+) {
+    using TR = VecTraits<T>;
+    using A = typename TR::acc_t;
+    constexpr int VEC = TR::VEC;
+    const long long nvec = n / VEC;
+    double sr = 0;
+    float srx = 0.f, sr2 = 0.f;
+    double srx_d = 0, sr2_d = 0;
+    A xv[VEC], yv[VEC];
+    long long lane_iters = 0;
+    for (long long i = gid; i < nvec; i += gstride) {
+        TR::load(x + i * VEC, xv);
+        TR::load(y + i * VEC, yv);
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) {
+            const double xd = (double)xv[j];
+            const double r = (double)yv[j] - (a + b * xd);
+            sr += r;
+            if constexpr (sizeof(A) == 8) {  // f64 inputs: full f64 products
+                srx_d += r * xd;
+                sr2_d += r * r;
+            } else {
+                const float r32 = (float)r;
+                srx += r32 * (float)xd;
+                sr2 += r32 * r32;
+            }
+        }
+        // bound the f32 partial span (keeps rounding growth ~sqrt(span))
+        if (((++lane_iters) & 255) == 0) {
+            srx_d += (double)srx;
+            sr2_d += (double)sr2;
+            srx = 0.f;
+            sr2 = 0.f;
+        }
+    }
+    for (long long i = nvec * VEC + gid; i < n; i += gstride) {
+        const double xi = (double)TR::get(x, i);
+        const double r = (double)TR::get(y, i) - (a + b * xi);
+        sr += r;
+        if constexpr (sizeof(A) == 8) {
+            srx_d += r * xi;
+            sr2_d += r * r;
+        } else {
+            const float r32 = (float)r;
+            srx += r32 * (float)xi;
+            sr2 += r32 * r32;
+        }
+    }
+    sr_out = sr;
+    srx_out = srx_d + (double)srx;
+    sr2_out = sr2_d + (double)sr2;
+}
+
 // Single-launch variant: pass1 + in-launch combine by the LAST-arriving
 // block (agent-scope release/acquire per cdna_hip_programming.md §6 G16;
 // saves the finish launch + its ~4.4 us single-block latency).  The ticket
@@ -176,29 +241,9 @@ __global__ __launch_bounds__(256) void k_gaussian_linear_fused(
 
     const long long gid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
     const long long gstride = (long long)gridDim.x * blockDim.x;
-    const long long nvec = n / VEC;
 
-    double sr = 0, srx = 0, sr2 = 0;
-    A xv[VEC], yv[VEC];
-    for (long long i = gid; i < nvec; i += gstride) {
-        TR::load(x + i * VEC, xv);
-        TR::load(y + i * VEC, yv);
-#pragma unroll
-        for (int j = 0; j < VEC; ++j) {
-            const double xd = (double)xv[j];
-            const double r = (double)yv[j] - (a + b * xd);
-            sr += r;
-            srx += r * xd;
-            sr2 += r * r;
-        }
-    }
-    for (long long i = nvec * VEC + gid; i < n; i += gstride) {
-        const double xi = (double)TR::get(x, i);
-        const double r = (double)TR::get(y, i) - (a + b * xi);
-        sr += r;
-        srx += r * xi;
-        sr2 += r * r;
-    }
+    double sr, srx, sr2;
+    gauss_accumulate<T>(x, y, n, gid, gstride, a, b, sr, srx, sr2);
 
     __shared__ double lds[4 * 3 + 1];  // ONE shared object (reduce + flag)
     double acc[3] = {sr2, sr, srx};
@@ -1247,27 +1292,8 @@ __global__ __launch_bounds__(256) void k_gaussian_persistent(
         my_seq += 1;
 
         // ---- compute this block's partials ------------------------------
-        double sr = 0, srx = 0, sr2 = 0;
-        A xv[VEC], yv[VEC];
-        for (long long i = gid; i < nvec; i += gstride) {
-            TR::load(x + i * VEC, xv);
-            TR::load(y + i * VEC, yv);
-#pragma unroll
-            for (int j = 0; j < VEC; ++j) {
-                const double xd = (double)xv[j];
-                const double r = (double)yv[j] - (a + b * xd);
-                sr += r;
-                srx += r * xd;
-                sr2 += r * r;
-            }
-        }
-        for (long long i = nvec * VEC + gid; i < n; i += gstride) {
-            const double xi = (double)TR::get(x, i);
-            const double r = (double)TR::get(y, i) - (a + b * xi);
-            sr += r;
-            srx += r * xi;
-            sr2 += r * r;
-        }
+        double sr, srx, sr2;
+        gauss_accumulate<T>(x, y, n, gid, gstride, a, b, sr, srx, sr2);
         PK_STAMP(3)
         double acc[3] = {sr2, sr, srx};
         block_reduce_add<3>(acc, lds);
