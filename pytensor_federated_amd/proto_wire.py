@@ -143,9 +143,13 @@ def decode_fields(buf: bytes) -> List[Tuple[int, int, object]]:
             fields.append((fnum, wtype, view[pos : pos + size]))
             pos += size
         elif wtype == 5:
+            if pos + 4 > n:
+                raise ValueError("truncated fixed32 field")
             fields.append((fnum, wtype, bytes(view[pos : pos + 4])))
             pos += 4
         elif wtype == 1:
+            if pos + 8 > n:
+                raise ValueError("truncated fixed64 field")
             fields.append((fnum, wtype, bytes(view[pos : pos + 8])))
             pos += 8
         else:
