@@ -57,11 +57,10 @@ def run_model(host: str, ports, parallel: bool, map_steps: int, draws: int):
             accepted += 1
         chain.append(theta.copy())
     chain = np.asarray(chain)
-    print(
-        f"Posterior over {draws} draws (accept {accepted / draws:.0%}): "
-        f"intercept={chain[:, 0].mean():.4f}+-{chain[:, 0].std():.4f} "
-        f"slope={chain[:, 1].mean():.4f}+-{chain[:, 1].std():.4f}"
-    )
+    from pytensor_federated_amd.inference import summary
+
+    print(f"Posterior over {draws} draws (accept {accepted / draws:.0%}):")
+    print(summary({"intercept": chain[:, 0], "slope": chain[:, 1]}))
     return chain
 
 

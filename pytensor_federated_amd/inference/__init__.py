@@ -8,6 +8,7 @@ nothing beyond this package: every driver consumes the framework's
 ``LogpGradServiceClient.evaluate``, or a ``FederatedShardEngine`` all plug
 in unchanged (the logp+grad of one MCMC step is ONE fused worker call).
 """
+from .diagnostics import effective_sample_size, split_rhat, summary  # noqa: F401
 from .mala import sample_mala_batched  # noqa: F401
 from .map import find_map  # noqa: F401
 from .mcmc import Metropolis, sample_metropolis  # noqa: F401

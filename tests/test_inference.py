@@ -197,3 +197,43 @@ def test_nuts_over_live_worker():
     finally:
         proc.terminate()
         proc.join(timeout=10)
+
+
+class TestDiagnostics:
+    def test_rhat_converged_vs_split_chains(self):
+        from pytensor_federated_amd.inference import split_rhat
+
+        rng = np.random.default_rng(11)
+        good = rng.standard_normal((4, 2000))
+        assert abs(split_rhat(good) - 1.0) < 0.02
+        # chains stuck at different means -> R-hat far above 1
+        bad = good + np.array([0.0, 0.0, 3.0, 3.0])[:, None]
+        assert split_rhat(bad) > 1.5
+
+    def test_ess_iid_vs_autocorrelated(self):
+        from pytensor_federated_amd.inference import effective_sample_size
+
+        rng = np.random.default_rng(12)
+        iid = rng.standard_normal((2, 4000))
+        ess_iid = effective_sample_size(iid)
+        assert ess_iid > 0.6 * 8000
+        # AR(1) with phi=0.95: ESS ~ N * (1-phi)/(1+phi) ~ 0.026 N
+        phi = 0.95
+        ar = np.empty((2, 4000))
+        for c in range(2):
+            e = rng.standard_normal(4000)
+            ar[c, 0] = e[0]
+            for t in range(1, 4000):
+                ar[c, t] = phi * ar[c, t - 1] + np.sqrt(1 - phi**2) * e[t]
+        ess_ar = effective_sample_size(ar)
+        assert ess_ar < 0.15 * 8000
+        assert ess_ar > 20
+
+    def test_summary_formats(self):
+        from pytensor_federated_amd.inference import summary
+
+        rng = np.random.default_rng(13)
+        text = summary({"intercept": rng.normal(1.5, 0.1, (2, 500)),
+                        "slope": rng.normal(0.5, 0.05, (2, 500))})
+        assert "intercept" in text and "r_hat" in text
+        assert len(text.splitlines()) == 3
