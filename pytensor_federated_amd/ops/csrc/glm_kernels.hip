@@ -978,18 +978,22 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched(
     constexpr int n_chunks = K / BL_CHUNK;
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    // Theta is staged PER CHUNK (not whole-K): keeps the block's LDS at
-    // ~46 KB -> 3 blocks/CU instead of 2 (the kernel was 60% wave-parked
-    // on chunk barriers; more co-resident blocks hide them).
+    unsigned short* th_lds = (unsigned short*)smem;                 // [BCH][K+TPAD]
+    const int th_stride = K + TPAD;
+    unsigned short* x_lds = th_lds + BCH * th_stride;               // [2][BL_ROWS][BL_CHUNK+XPAD]
     const int x_stride = BL_CHUNK + XPAD;
     const int x_buf = BL_ROWS * x_stride;
-    const int th_buf = BCH * x_stride;
-    unsigned short* x_lds = (unsigned short*)smem;                  // [2][BL_ROWS][BL_CHUNK+XPAD]
-    unsigned short* th_lds = x_lds + 2 * x_buf;                     // [2][BCH][BL_CHUNK+XPAD]
-    unsigned short* rt_lds = th_lds + 2 * th_buf;                   // [BCH][BL_ROWS+RPAD]
+    unsigned short* rt_lds = x_lds + 2 * x_buf;                     // [BCH][BL_ROWS+RPAD]
     const int rt_stride = BL_ROWS + RPAD;
     float* y_lds = (float*)(rt_lds + BCH * rt_stride + 8);          // [BL_ROWS]
     float* red_lds = y_lds + BL_ROWS;                               // [256]
+
+    // ---- stage Theta^T once per block ----
+    for (int idx = threadIdx.x * 8; idx < BCH * K; idx += 256 * 8) {
+        const int b = idx / K;
+        const int k = idx % K;
+        *(U4*)&th_lds[b * th_stride + k] = *(const U4*)&theta_t[b * K + k];
+    }
 
     typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
     typedef __attribute__((ext_vector_type(4))) float f32x4_t;
@@ -1019,42 +1023,27 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched(
 #define LOAD_CHUNK(c)                                                              _Pragma("unroll") for (int rr = 0; rr < 4; ++rr) {                                 const long long row = row0 + st_r0 + rr * 16;                                  ld[rr] = (U4){0, 0, 0, 0};                                                     if (row < n_rows)                                                                  ld[rr] = *(const U4*)&X[row * (long long)K + (c) * BL_CHUNK + st_k0];     }
 #define WRITE_CHUNK(buf)                                                           _Pragma("unroll") for (int rr = 0; rr < 4; ++rr)                                   *(U4*)&x_lds[(buf) * x_buf + (st_r0 + rr * 16) * x_stride + st_k0] = ld[rr];
 
-#define LOAD_TH(c)                                                             \
-    th_ld = *(const U4*)&theta_t[(threadIdx.x / 16) * K + (c) * BL_CHUNK       \
-                                 + (threadIdx.x % 16) * 8];
-#define WRITE_TH(buf)                                                          \
-    *(U4*)&th_lds[(buf) * th_buf + (threadIdx.x / 16) * x_stride               \
-                  + (threadIdx.x % 16) * 8] = th_ld;
-
-        U4 th_ld;
         LOAD_CHUNK(0)
-        LOAD_TH(0)
         WRITE_CHUNK(0)
-        WRITE_TH(0)
         int cur = 0;
 
         // ---- phase A: Z = X . Theta ----
         f32x4_t z_acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int c = 0; c < n_chunks; ++c) {
-            __syncthreads();  // buf[cur] visible
-            if (c + 1 < n_chunks) {
-                LOAD_CHUNK(c + 1)
-                LOAD_TH(c + 1)
-            }
+            __syncthreads();  // buf[cur] (and Theta on c==0) visible
+            if (c + 1 < n_chunks) LOAD_CHUNK(c + 1)
 #pragma unroll
             for (int ks = 0; ks < BL_CHUNK / 32; ++ks) {
                 frag_u a, b;
                 const int arow = wid * 16 + (lane & 15);
                 const int ak = ks * 32 + (lane >> 4) * 8;
                 a.q = *(U4*)&x_lds[cur * x_buf + arow * x_stride + ak];
-                b.q = *(U4*)&th_lds[cur * th_buf + (lane & 15) * x_stride + ak];
+                const int bk = c * BL_CHUNK + ks * 32 + (lane >> 4) * 8;
+                b.q = *(U4*)&th_lds[(lane & 15) * th_stride + bk];
                 z_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, z_acc, 0, 0, 0);
             }
-            if (c + 1 < n_chunks) {
-                WRITE_CHUNK(cur ^ 1)
-                WRITE_TH(cur ^ 1)
-            }
+            if (c + 1 < n_chunks) WRITE_CHUNK(cur ^ 1)
             cur ^= 1;
         }
 
@@ -1111,8 +1100,6 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched(
         __syncthreads();  // rt_lds reuse next tile
 #undef LOAD_CHUNK
 #undef WRITE_CHUNK
-#undef LOAD_TH
-#undef WRITE_TH
     }
 
     // ---- epilogue: block partials -> slab ----
@@ -1156,7 +1143,7 @@ extern "C" int fed_logistic_glm_batched(
         grid = (int)(ws_bytes / (slab_cols * 4));
     if (grid < 1) return -3;
     const int lds_bytes =
-        (2 * BL_ROWS * (BL_CHUNK + XPAD) + 2 * BCH * (BL_CHUNK + XPAD) + BCH * (BL_ROWS + RPAD) + 8) * 2 +
+        (BCH * (K + TPAD) + 2 * BL_ROWS * (BL_CHUNK + XPAD) + BCH * (BL_ROWS + RPAD) + 8) * 2 +
         (BL_ROWS + 256) * 4 + 64;
     if (K == 1024)
         hipLaunchKernelGGL(k_logistic_glm_batched<1024>, dim3(grid), dim3(block), lds_bytes,
