@@ -73,6 +73,19 @@ class LogisticGLMModel(LogpGradModel):
         if self._X.dim() != 2 or self._y.dim() != 1 or self._X.shape[0] != self._y.shape[0]:
             raise ValueError("X must be [N,K] and y [N].")
         self._n, self._k = self._X.shape
+        self._k_pad = 0
+        if (use_kernels is None or use_kernels) and self._X.is_cuda:
+            # pad the feature dim to the fused kernel's lane-slice granule
+            # (512 bf16 / 256 f32 per wave64 pass); zero columns change
+            # nothing: z is unaffected and their gradient is exactly 0.
+            granule = 512 if dtype == torch.bfloat16 else 256
+            rem = self._k % granule
+            if rem:
+                self._k_pad = granule - rem
+                pad = torch.zeros(
+                    (self._n, self._k_pad), device=self._X.device, dtype=self._X.dtype
+                )
+                self._X = torch.cat([self._X, pad], dim=1).contiguous()
 
     @property
     def device(self):
@@ -86,18 +99,23 @@ class LogisticGLMModel(LogpGradModel):
     def n_features(self) -> int:
         return int(self._k)
 
+    @property
+    def _k_eff(self) -> int:
+        """Feature dim as the kernel sees it (incl. zero padding)."""
+        return self._k + self._k_pad
+
     def _kernel_path(self) -> bool:
         want = self._X.is_cuda if self._use_kernels is None else self._use_kernels
         if want and self._use_kernels is None:
             from ..ops import logistic_kernel_supports
 
-            if not logistic_kernel_supports(self._X.dtype, self._k):
+            if not logistic_kernel_supports(self._X.dtype, self._k_eff):
                 import warnings
 
                 warnings.warn(
                     f"fused logistic kernel not compiled for (dtype={self._X.dtype}, "
-                    f"K={self._k}); using the eager two-matmul path (~2x slower). "
-                    f"Supported K: multiples of 512 (bf16) / 256 (f32).",
+                    f"K={self._k_eff}); using the eager two-matmul path (~2x slower). "
+                    f"Supported padded K: up to 2048 (bf16) / 1024 (f32).",
                     stacklevel=3,
                 )
                 return False
@@ -117,6 +135,21 @@ class LogisticGLMModel(LogpGradModel):
         if self._kernel_path():
             from ..ops import logistic_glm_logp_grad
 
+            if self._k_pad:
+                beta = torch.cat(
+                    [beta.to(torch.float32),
+                     torch.zeros(self._k_pad, dtype=torch.float32, device=beta.device)]
+                )
+                # ``out`` (if given) is sized [1 + k]; the kernel writes
+                # [1 + k_eff] -- evaluate into the kernel's own buffer and
+                # copy the un-padded slice over.
+                logp, grad = logistic_glm_logp_grad(self._X, self._y, beta)
+                grad = grad[: self._k]
+                if out is not None:
+                    out[0] = logp
+                    out[1:] = grad
+                    return out[0], [out[1:]]
+                return logp, [grad]
             logp, grad = logistic_glm_logp_grad(self._X, self._y, beta, out=out)
             return logp, [grad]
         logp, grads = self._logp_grad_eager(beta)
@@ -139,15 +172,22 @@ class LogisticGLMModel(LogpGradModel):
             self._kernel_path()
             and self._X.dtype == torch.bfloat16
             and theta.shape[1] == 16
-            and self._k in (512, 1024)
+            and self._k_eff in (512, 1024)
         ):
             from ..ops import logistic_glm_logp_grad_batched
 
-            return logistic_glm_logp_grad_batched(self._X, self._y, theta)
+            if self._k_pad:
+                theta = torch.cat(
+                    [theta.to(torch.float32),
+                     torch.zeros(self._k_pad, theta.shape[1], dtype=torch.float32,
+                                 device=theta.device)]
+                )
+            logp, G = logistic_glm_logp_grad_batched(self._X, self._y, theta)
+            return logp, (G[: self._k] if self._k_pad else G)
         return self._logp_grad_batched_eager(theta)
 
     def _logp_grad_batched_eager(self, theta: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
-        X, y = self._X, self._y
+        X, y = self._X[:, : self._k], self._y
         acc_dtype = torch.float64 if X.dtype == torch.float64 else torch.float32
         theta = theta.to(device=X.device, dtype=acc_dtype)
         Xf = X.to(acc_dtype)
@@ -161,7 +201,7 @@ class LogisticGLMModel(LogpGradModel):
         return logp, G
 
     def _logp_grad_eager(self, beta: torch.Tensor) -> Tuple[torch.Tensor, List[torch.Tensor]]:
-        X, y = self._X, self._y
+        X, y = self._X[:, : self._k], self._y
         acc_dtype = torch.float64 if X.dtype == torch.float64 else torch.float32
         beta = beta.to(device=X.device, dtype=acc_dtype)
         Xf = X.to(acc_dtype)

@@ -99,11 +99,23 @@ class TestLogisticKernel:
             g_k.cpu().numpy(), g_e.cpu().numpy(), rtol=rtol, atol=rtol * 10
         )
 
-    def test_rejects_unsupported_k(self, dev):
-        X, y, beta0 = generate_logistic_dataset(100, 100, seed=24)
+    def test_unaligned_k_pads_and_matches_eager(self, dev):
+        # K=100 pads to the 512 lane-slice granule; zero columns are inert
+        X, y, beta0 = generate_logistic_dataset(5000, 100, seed=24)
         m = LogisticGLMModel(X, y, device=dev, dtype=torch.bfloat16, use_kernels=True)
-        with pytest.raises(RuntimeError):
-            m.logp_grad(torch.as_tensor(beta0, dtype=torch.float32))
+        assert m._k_pad == 412 and m._X.shape[1] == 512
+        beta32 = torch.as_tensor(beta0, dtype=torch.float32)
+        logp_k, (g_k,) = m.logp_grad(beta32)
+        eager = LogisticGLMModel(
+            m._X[:, :100].contiguous(), m._y, device=dev, dtype=torch.bfloat16,
+            use_kernels=False,
+        )
+        logp_e, (g_e,) = eager.logp_grad(beta32)
+        assert g_k.shape == (100,)
+        np.testing.assert_allclose(float(logp_k), float(logp_e), rtol=2e-4)
+        np.testing.assert_allclose(
+            g_k.cpu().numpy(), g_e.cpu().numpy(), rtol=2e-4, atol=2e-3
+        )
 
 
 class TestStreamsOnGPU:
