@@ -1126,6 +1126,152 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched(
     }
 }
 
+
+// ---------------------------------------------------------------------------
+// Batched logistic, LDS-resident-tile variant: phase B never re-reads HBM
+// ---------------------------------------------------------------------------
+// The chunked variant's phase-B re-read misses L2 (2 blocks/CU x 32 CUs x
+// 128 KB tiles = 8 MB active per 4 MB XCD L2) -> ~2x HBM traffic.  Here ONE
+// block per CU holds its whole [64][K] tile in LDS (132 KB at K=1024),
+// stages it once with a deep load pipeline, and runs both MFMA phases out
+// of LDS.  Theta fragments read straight from L2 (32 KB, resident).
+
+template <int K>
+__global__ __launch_bounds__(256, 1) void k_logistic_glm_batched_lds(
+    const unsigned short* __restrict__ X,   // [N][K] bf16
+    const unsigned short* __restrict__ y,   // [N] bf16
+    long long n_rows,
+    const unsigned short* __restrict__ theta_t,  // [BCH][K] bf16
+    float* __restrict__ slab                     // [grid][BCH + K*BCH]
+) {
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    constexpr int n_chunks = K / BL_CHUNK;
+    constexpr int x_stride = K + XPAD;
+
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    unsigned short* x_lds = (unsigned short*)smem;              // [BL_ROWS][K+XPAD]
+    unsigned short* rt_lds = x_lds + BL_ROWS * x_stride;        // [BCH][BL_ROWS+RPAD]
+    const int rt_stride = BL_ROWS + RPAD;
+    float* y_lds = (float*)(rt_lds + BCH * rt_stride + 8);      // [BL_ROWS]
+    float* red_lds = y_lds + BL_ROWS;                           // [256]
+
+    typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
+    typedef __attribute__((ext_vector_type(4))) float f32x4_t;
+    union frag_u { bf16x8_t v; unsigned short u[8]; U4 q; };
+
+    f32x4_t g_acc[n_chunks * 2];
+#pragma unroll
+    for (int t = 0; t < n_chunks * 2; ++t) g_acc[t] = (f32x4_t){0.f, 0.f, 0.f, 0.f};
+    float logp_acc = 0.f;
+
+    const long long n_tiles = (n_rows + BL_ROWS - 1) / BL_ROWS;
+    for (long long tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
+        const long long row0 = tile * BL_ROWS;
+        if (threadIdx.x < BL_ROWS) {
+            const long long r = row0 + threadIdx.x;
+            y_lds[threadIdx.x] = r < n_rows ? bf16_bits_to_f32(y[r]) : 0.f;
+        }
+        // ---- stage the WHOLE tile: 512 B per thread, 32 deep loads ----
+        {
+            const int r = threadIdx.x / 16;          // owns rows r, r+16, ...
+            const int k0 = (threadIdx.x % 16) * 8;   // k column start
+#pragma unroll
+            for (int rr = 0; rr < 4; ++rr) {
+                const long long row = row0 + r + rr * 16;
+#pragma unroll
+                for (int kk = 0; kk < K / 128; ++kk) {
+                    U4 val = {0, 0, 0, 0};
+                    if (row < n_rows)
+                        val = *(const U4*)&X[row * (long long)K + kk * 128 + k0];
+                    *(U4*)&x_lds[(r + rr * 16) * x_stride + kk * 128 + k0] = val;
+                }
+            }
+        }
+        __syncthreads();
+
+        // ---- phase A: Z (LDS x, L2 theta) ----
+        f32x4_t z_acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int ks = 0; ks < K / 32; ++ks) {
+            frag_u a, b;
+            const int arow = wid * 16 + (lane & 15);
+            const int ak = ks * 32 + (lane >> 4) * 8;
+            a.q = *(U4*)&x_lds[arow * x_stride + ak];
+            b.q = *(const U4*)&theta_t[(lane & 15) * K + ak];
+            z_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, z_acc, 0, 0, 0);
+        }
+
+        // ---- logp + R ----
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int row_in_wave = (lane >> 4) * 4 + r;
+            const int row_in_tile = wid * 16 + row_in_wave;
+            const long long row = row0 + row_in_tile;
+            const int chain = lane & 15;
+            float z = z_acc[r];
+            float yv = y_lds[row_in_tile];
+            float resid = 0.f;
+            if (row < n_rows) {
+                const float sp = fmaxf(z, 0.f) + log1pf(__expf(-fabsf(z)));
+                logp_acc += yv * z - sp;
+                resid = yv - 1.f / (1.f + __expf(-z));
+            }
+            union { float f; unsigned int u; } cv;
+            cv.f = resid;
+            const unsigned int rnd = 0x7fff + ((cv.u >> 16) & 1);
+            rt_lds[chain * rt_stride + row_in_tile] = (unsigned short)((cv.u + rnd) >> 16);
+        }
+        __syncthreads();  // R complete
+
+        // ---- phase B: G += X^T R (all LDS) ----
+#pragma unroll
+        for (int c = 0; c < n_chunks; ++c) {
+#pragma unroll
+            for (int t2 = 0; t2 < 2; ++t2) {
+                const int kcol0 = c * BL_CHUNK + wid * 32 + t2 * 16;
+                f32x4_t acc = g_acc[c * 2 + t2];
+#pragma unroll
+                for (int rs = 0; rs < 2; ++rs) {
+                    frag_u a, b;
+                    const int kcol = kcol0 + (lane & 15);
+                    const int arow0 = rs * 32 + (lane >> 4) * 8;
+#pragma unroll
+                    for (int j = 0; j < 8; ++j)
+                        a.u[j] = x_lds[(arow0 + j) * x_stride + kcol];
+                    b.q = *(U4*)&rt_lds[(lane & 15) * rt_stride + rs * 32 + (lane >> 4) * 8];
+                    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, acc, 0, 0, 0);
+                }
+                g_acc[c * 2 + t2] = acc;
+            }
+        }
+        __syncthreads();  // x_lds/rt_lds reuse next tile
+    }
+
+    // ---- epilogue (same slab layout as the chunked variant) ----
+    red_lds[threadIdx.x] = logp_acc;
+    __syncthreads();
+    float* slab_blk = slab + (long long)blockIdx.x * (BCH + (long long)K * BCH);
+    if (threadIdx.x < BCH) {
+        float s = 0.f;
+        for (int i = threadIdx.x; i < 256; i += BCH) s += red_lds[i];
+        slab_blk[threadIdx.x] = s;
+    }
+    float* g_slab = slab_blk + BCH;
+#pragma unroll
+    for (int c = 0; c < n_chunks; ++c) {
+#pragma unroll
+        for (int t2 = 0; t2 < 2; ++t2) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int kcol = c * BL_CHUNK + wid * 32 + t2 * 16 + (lane >> 4) * 4 + r;
+                const int chain = lane & 15;
+                g_slab[(long long)kcol * BCH + chain] = g_acc[c * 2 + t2][r];
+            }
+        }
+    }
+}
+
 extern "C" int fed_logistic_glm_batched(
     const void* X, const void* y, long long n_rows, int K,
     const void* theta_t_bf16,  // [16][K] bf16, transposed theta
@@ -1517,148 +1663,3 @@ int fed_gaussian_persistent_stop(void* handle) {
 }
 
 }  // extern "C"
-
-// ---------------------------------------------------------------------------
-// Batched logistic, LDS-resident-tile variant: phase B never re-reads HBM
-// ---------------------------------------------------------------------------
-// The chunked variant's phase-B re-read misses L2 (2 blocks/CU x 32 CUs x
-// 128 KB tiles = 8 MB active per 4 MB XCD L2) -> ~2x HBM traffic.  Here ONE
-// block per CU holds its whole [64][K] tile in LDS (132 KB at K=1024),
-// stages it once with a deep load pipeline, and runs both MFMA phases out
-// of LDS.  Theta fragments read straight from L2 (32 KB, resident).
-
-template <int K>
-__global__ __launch_bounds__(256, 1) void k_logistic_glm_batched_lds(
-    const unsigned short* __restrict__ X,   // [N][K] bf16
-    const unsigned short* __restrict__ y,   // [N] bf16
-    long long n_rows,
-    const unsigned short* __restrict__ theta_t,  // [BCH][K] bf16
-    float* __restrict__ slab                     // [grid][BCH + K*BCH]
-) {
-    const int lane = threadIdx.x & 63;
-    const int wid = threadIdx.x >> 6;
-    constexpr int n_chunks = K / BL_CHUNK;
-    constexpr int x_stride = K + XPAD;
-
-    extern __shared__ __attribute__((aligned(16))) char smem[];
-    unsigned short* x_lds = (unsigned short*)smem;              // [BL_ROWS][K+XPAD]
-    unsigned short* rt_lds = x_lds + BL_ROWS * x_stride;        // [BCH][BL_ROWS+RPAD]
-    const int rt_stride = BL_ROWS + RPAD;
-    float* y_lds = (float*)(rt_lds + BCH * rt_stride + 8);      // [BL_ROWS]
-    float* red_lds = y_lds + BL_ROWS;                           // [256]
-
-    typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
-    typedef __attribute__((ext_vector_type(4))) float f32x4_t;
-    union frag_u { bf16x8_t v; unsigned short u[8]; U4 q; };
-
-    f32x4_t g_acc[n_chunks * 2];
-#pragma unroll
-    for (int t = 0; t < n_chunks * 2; ++t) g_acc[t] = (f32x4_t){0.f, 0.f, 0.f, 0.f};
-    float logp_acc = 0.f;
-
-    const long long n_tiles = (n_rows + BL_ROWS - 1) / BL_ROWS;
-    for (long long tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
-        const long long row0 = tile * BL_ROWS;
-        if (threadIdx.x < BL_ROWS) {
-            const long long r = row0 + threadIdx.x;
-            y_lds[threadIdx.x] = r < n_rows ? bf16_bits_to_f32(y[r]) : 0.f;
-        }
-        // ---- stage the WHOLE tile: 512 B per thread, 32 deep loads ----
-        {
-            const int r = threadIdx.x / 16;          // owns rows r, r+16, ...
-            const int k0 = (threadIdx.x % 16) * 8;   // k column start
-#pragma unroll
-            for (int rr = 0; rr < 4; ++rr) {
-                const long long row = row0 + r + rr * 16;
-#pragma unroll
-                for (int kk = 0; kk < K / 128; ++kk) {
-                    U4 val = {0, 0, 0, 0};
-                    if (row < n_rows)
-                        val = *(const U4*)&X[row * (long long)K + kk * 128 + k0];
-                    *(U4*)&x_lds[(r + rr * 16) * x_stride + kk * 128 + k0] = val;
-                }
-            }
-        }
-        __syncthreads();
-
-        // ---- phase A: Z (LDS x, L2 theta) ----
-        f32x4_t z_acc = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-        for (int ks = 0; ks < K / 32; ++ks) {
-            frag_u a, b;
-            const int arow = wid * 16 + (lane & 15);
-            const int ak = ks * 32 + (lane >> 4) * 8;
-            a.q = *(U4*)&x_lds[arow * x_stride + ak];
-            b.q = *(const U4*)&theta_t[(lane & 15) * K + ak];
-            z_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, z_acc, 0, 0, 0);
-        }
-
-        // ---- logp + R ----
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            const int row_in_wave = (lane >> 4) * 4 + r;
-            const int row_in_tile = wid * 16 + row_in_wave;
-            const long long row = row0 + row_in_tile;
-            const int chain = lane & 15;
-            float z = z_acc[r];
-            float yv = y_lds[row_in_tile];
-            float resid = 0.f;
-            if (row < n_rows) {
-                const float sp = fmaxf(z, 0.f) + log1pf(__expf(-fabsf(z)));
-                logp_acc += yv * z - sp;
-                resid = yv - 1.f / (1.f + __expf(-z));
-            }
-            union { float f; unsigned int u; } cv;
-            cv.f = resid;
-            const unsigned int rnd = 0x7fff + ((cv.u >> 16) & 1);
-            rt_lds[chain * rt_stride + row_in_tile] = (unsigned short)((cv.u + rnd) >> 16);
-        }
-        __syncthreads();  // R complete
-
-        // ---- phase B: G += X^T R (all LDS) ----
-#pragma unroll
-        for (int c = 0; c < n_chunks; ++c) {
-#pragma unroll
-            for (int t2 = 0; t2 < 2; ++t2) {
-                const int kcol0 = c * BL_CHUNK + wid * 32 + t2 * 16;
-                f32x4_t acc = g_acc[c * 2 + t2];
-#pragma unroll
-                for (int rs = 0; rs < 2; ++rs) {
-                    frag_u a, b;
-                    const int kcol = kcol0 + (lane & 15);
-                    const int arow0 = rs * 32 + (lane >> 4) * 8;
-#pragma unroll
-                    for (int j = 0; j < 8; ++j)
-                        a.u[j] = x_lds[(arow0 + j) * x_stride + kcol];
-                    b.q = *(U4*)&rt_lds[(lane & 15) * rt_stride + rs * 32 + (lane >> 4) * 8];
-                    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, acc, 0, 0, 0);
-                }
-                g_acc[c * 2 + t2] = acc;
-            }
-        }
-        __syncthreads();  // x_lds/rt_lds reuse next tile
-    }
-
-    // ---- epilogue (same slab layout as the chunked variant) ----
-    red_lds[threadIdx.x] = logp_acc;
-    __syncthreads();
-    float* slab_blk = slab + (long long)blockIdx.x * (BCH + (long long)K * BCH);
-    if (threadIdx.x < BCH) {
-        float s = 0.f;
-        for (int i = threadIdx.x; i < 256; i += BCH) s += red_lds[i];
-        slab_blk[threadIdx.x] = s;
-    }
-    float* g_slab = slab_blk + BCH;
-#pragma unroll
-    for (int c = 0; c < n_chunks; ++c) {
-#pragma unroll
-        for (int t2 = 0; t2 < 2; ++t2) {
-#pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                const int kcol = c * BL_CHUNK + wid * 32 + t2 * 16 + (lane >> 4) * 4 + r;
-                const int chain = lane & 15;
-                g_slab[(long long)kcol * BCH + chain] = g_acc[c * 2 + t2][r];
-            }
-        }
-    }
-}
