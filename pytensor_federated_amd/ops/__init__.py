@@ -298,6 +298,7 @@ class PersistentLinearEngine:
         self._dtype_code = _DTYPE_CODE[x.dtype]
         self._out = (ctypes.c_double * 3)()
         self._handle = None
+        self._lock = threading.Lock()  # one mailbox -> one eval at a time
         self._start()
 
     def _start(self) -> None:
@@ -309,20 +310,21 @@ class PersistentLinearEngine:
             raise RuntimeError("fed_gaussian_persistent_start failed")
 
     def logp_grad_sync(self, a: float, b: float) -> Tuple[float, float, float]:
-        rc = self._lib.fed_gaussian_persistent_eval(
-            self._handle, float(a), float(b), self._out
-        )
-        if rc == -6:
-            # server self-exited (idle give-up); relaunch once
-            self._lib.fed_gaussian_persistent_stop(self._handle)
-            self._handle = None
-            self._start()
+        with self._lock:
             rc = self._lib.fed_gaussian_persistent_eval(
                 self._handle, float(a), float(b), self._out
             )
-        if rc != 0:
-            raise RuntimeError(f"persistent eval failed ({rc})")
-        return self._out[0], self._out[1], self._out[2]
+            if rc == -6:
+                # server unreachable; full restart once
+                self._lib.fed_gaussian_persistent_stop(self._handle)
+                self._handle = None
+                self._start()
+                rc = self._lib.fed_gaussian_persistent_eval(
+                    self._handle, float(a), float(b), self._out
+                )
+            if rc != 0:
+                raise RuntimeError(f"persistent eval failed ({rc})")
+            return self._out[0], self._out[1], self._out[2]
 
     def close(self) -> None:
         if getattr(self, "_handle", None):
