@@ -38,6 +38,7 @@ __device__ __forceinline__ float bf16_bits_to_f32(unsigned short u) {
 // 16-byte vector loads: 8 bf16 / 4 f32 / 2 f64 per lane per instruction
 // (G13: hipcc does not auto-vectorize bf16 loads; scalar bf16 is ~2x slower).
 struct U4 { unsigned int x, y, z, w; };
+typedef __attribute__((ext_vector_type(4))) unsigned int u32x4_t;
 
 template <typename T> struct VecTraits;
 template <> struct VecTraits<float> {
@@ -48,6 +49,14 @@ template <> struct VecTraits<float> {
         out[0] = v.x; out[1] = v.y; out[2] = v.z; out[3] = v.w;
     }
     __device__ static inline float get(const float* p, long long i) { return p[i]; }
+    __device__ static inline void unpack(const u32x4_t v, float* out) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            union { unsigned int u; float f; } c;
+            c.u = v[j];
+            out[j] = c.f;
+        }
+    }
 };
 template <> struct VecTraits<double> {
     static constexpr int VEC = 2;
@@ -73,6 +82,13 @@ template <> struct VecTraits<bf16_tag> {
     }
     __device__ static inline float get(const bf16_tag* p, long long i) {
         return bf16_bits_to_f32(p[i].bits);
+    }
+    __device__ static inline void unpack(const u32x4_t v, float* out) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            out[2 * j]     = bf16_bits_to_f32((unsigned short)(v[j] & 0xffffu));
+            out[2 * j + 1] = bf16_bits_to_f32((unsigned short)(v[j] >> 16));
+        }
     }
 };
 
@@ -361,8 +377,14 @@ __global__ __launch_bounds__(256) void k_logistic_glm_reg(
         float z0p = 0.f, z1p = 0.f;
 #pragma unroll
         for (int c = 0; c < KITER; ++c) {
-            TR::load(row0 + c * WAVE * VEC + lane * VEC, xreg[c]);
-            TR::load(row1 + c * WAVE * VEC + lane * VEC, xreg2[c]);
+            // nontemporal: each X row is consumed once (from registers for
+            // both logp and grad) -- do not displace L2/L3 lines
+            const u32x4_t v0 = __builtin_nontemporal_load(
+                (const u32x4_t*)(row0 + c * WAVE * VEC + lane * VEC));
+            const u32x4_t v1 = __builtin_nontemporal_load(
+                (const u32x4_t*)(row1 + c * WAVE * VEC + lane * VEC));
+            TR::unpack(v0, xreg[c]);
+            TR::unpack(v1, xreg2[c]);
         }
 #pragma unroll
         for (int c = 0; c < KITER; ++c)
