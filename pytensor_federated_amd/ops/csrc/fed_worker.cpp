@@ -84,6 +84,21 @@ static std::string encode_f64_scalar(double value) {
     return msg;
 }
 
+// encode a 1-d float64 ndarray message (packed shape/strides fields)
+static std::string encode_f64_vector(const double* values, int n) {
+    std::string msg;
+    std::string data((const char*)values, (size_t)n * 8);
+    put_len_field(msg, 1, data);
+    put_len_field(msg, 2, "float64");
+    std::string shape;
+    put_varint(shape, (unsigned long long)n);
+    put_len_field(msg, 3, shape);           // packed repeated int64
+    std::string strides;
+    put_varint(strides, 8);
+    put_len_field(msg, 4, strides);
+    return msg;
+}
+
 struct ParsedArray {
     std::vector<unsigned char> data;
     std::string dtype;
@@ -165,7 +180,10 @@ static double scalar_value(const ParsedArray& arr) {
 typedef int (*eval_fn_t)(const void*, const void*, long long, double, double,
                          double, double*, double*, double*, long long, int, void*,
                          unsigned long long);
+typedef int (*logistic_fn_t)(const void*, const void*, long long, int,
+                             const float*, double*, float*, long long, int, void*);
 typedef void* (*host_alloc_fn_t)(long long);
+typedef void* (*dev_alloc_fn_t)(long long);
 
 enum { FED_F32 = 0, FED_F64 = 1, FED_BF16 = 2 };
 
@@ -173,12 +191,17 @@ struct Worker {
     void* x_dev = nullptr;
     void* y_dev = nullptr;
     long long n = 0;
+    int K = 0;                    // 0 = gaussian linear; >0 = logistic [N][K]
     double sigma = 0.4;
     int dtype = FED_BF16;
-    double* out_dev = nullptr;
+    double* out_dev = nullptr;    // fp64[3] or fp64[1+K]
+    float* beta_dev = nullptr;    // logistic: f32[K]
     double* ws_dev = nullptr;
-    double* mailbox = nullptr;
+    float* ws_f32 = nullptr;      // logistic grad slab
+    double* mailbox = nullptr;    // gaussian result path
+    std::vector<double> out_host; // logistic result readback
     eval_fn_t eval = nullptr;
+    logistic_fn_t eval_logistic = nullptr;
     unsigned long long seq = 0;
     int n_clients = 0;
 };
@@ -189,6 +212,42 @@ static unsigned short f32_to_bf16(float f) {
     // round-to-nearest-even like torch's float->bf16 conversion
     const unsigned int rounding = 0x7fff + ((u >> 16) & 1);
     return (unsigned short)((u + rounding) >> 16);
+}
+
+static bool load_logistic_shard(Worker& w, const char* path) {
+    FILE* f = fopen(path, "rb");
+    if (!f) { fprintf(stderr, "cannot open %s\n", path); return false; }
+    long long n = 0, K = 0;
+    if (fread(&n, 8, 1, f) != 1 || fread(&K, 8, 1, f) != 1 || n <= 0 ||
+        (K != 512 && K != 1024 && K != 2048)) {
+        fclose(f);
+        fprintf(stderr, "bad logistic shard header (K must be 512/1024/2048)\n");
+        return false;
+    }
+    w.n = n;
+    w.K = (int)K;
+    std::vector<double> buf(n * K);
+    std::vector<unsigned char> xb(n * K * 2), yb(n * 2);
+    if (fread(buf.data(), 8, n * K, f) != (size_t)(n * K)) { fclose(f); return false; }
+    for (long long i = 0; i < n * K; ++i) {
+        unsigned short v = f32_to_bf16((float)buf[i]);
+        memcpy(xb.data() + i * 2, &v, 2);
+    }
+    buf.resize(n);
+    if (fread(buf.data(), 8, n, f) != (size_t)n) { fclose(f); return false; }
+    fclose(f);
+    for (long long i = 0; i < n; ++i) {
+        unsigned short v = f32_to_bf16((float)buf[i]);
+        memcpy(yb.data() + i * 2, &v, 2);
+    }
+    if (hipMalloc(&w.x_dev, n * K * 2) != hipSuccess ||
+        hipMalloc(&w.y_dev, n * 2) != hipSuccess ||
+        hipMemcpy(w.x_dev, xb.data(), n * K * 2, hipMemcpyHostToDevice) != hipSuccess ||
+        hipMemcpy(w.y_dev, yb.data(), n * 2, hipMemcpyHostToDevice) != hipSuccess) {
+        fprintf(stderr, "device upload failed\n");
+        return false;
+    }
+    return true;
 }
 
 static bool load_shard(Worker& w, const char* path) {
@@ -284,7 +343,7 @@ static void serve_client(Worker& w, int fd) {
         payload.resize(ln);
         if (ln && !read_exact(fd, payload.data(), ln)) break;
         static std::mutex eval_mu;
-        if (hdr[0] == 0x01) {  // Evaluate
+        if (hdr[0] == 0x01 && w.K == 0) {  // Evaluate: gaussian linear
             std::vector<ParsedArray> items;
             std::string uuid;
             if (!parse_input_arrays(payload.data(), ln, items, uuid) || items.size() != 2) {
@@ -314,6 +373,44 @@ static void serve_client(Worker& w, int fd) {
             put_len_field(out, 1, encode_f64_scalar(res[2]));  // d/db
             put_len_field(out, 2, uuid);
             if (!write_frame(fd, 0x81, out)) break;
+        } else if (hdr[0] == 0x01) {  // Evaluate: logistic GLM, beta[K] f64 in
+            std::vector<ParsedArray> items;
+            std::string uuid;
+            if (!parse_input_arrays(payload.data(), ln, items, uuid) || items.size() != 1 ||
+                items[0].dtype != "float64" ||
+                items[0].data.size() != (size_t)w.K * 8) {
+                write_frame(fd, 0xFF, "expected one float64 beta[K] input");
+                continue;
+            }
+            std::string out;
+            {
+                std::lock_guard<std::mutex> lock(eval_mu);
+                std::vector<float> beta32(w.K);
+                const double* bd = (const double*)items[0].data.data();
+                for (int i = 0; i < w.K; ++i) beta32[i] = (float)bd[i];
+                if (hipMemcpy(w.beta_dev, beta32.data(), w.K * 4,
+                              hipMemcpyHostToDevice) != hipSuccess) {
+                    write_frame(fd, 0xFF, "beta upload failed");
+                    continue;
+                }
+                int rc = w.eval_logistic(w.x_dev, w.y_dev, w.n, w.K, w.beta_dev,
+                                         w.out_dev, w.ws_f32,
+                                         (long long)1024 * w.K * 4, w.dtype, nullptr);
+                if (rc == 0) rc = (int)hipDeviceSynchronize();
+                if (rc != 0) {
+                    write_frame(fd, 0xFF, "logistic eval failed");
+                    continue;
+                }
+                if (hipMemcpy(w.out_host.data(), w.out_dev, (1 + w.K) * 8,
+                              hipMemcpyDeviceToHost) != hipSuccess) {
+                    write_frame(fd, 0xFF, "result readback failed");
+                    continue;
+                }
+                put_len_field(out, 1, encode_f64_scalar(w.out_host[0]));          // logp
+                put_len_field(out, 1, encode_f64_vector(&w.out_host[1], w.K));    // grad
+            }
+            put_len_field(out, 2, uuid);
+            if (!write_frame(fd, 0x81, out)) break;
         } else if (hdr[0] == 0x02) {  // GetLoad
             std::string out;
             put_varint(out, (1 << 3) | 0);  // n_clients
@@ -331,11 +428,13 @@ static void serve_client(Worker& w, int fd) {
 int main(int argc, char** argv) {
     int port = 9600;
     const char* data_path = nullptr;
+    bool logistic = false;
     Worker w;
     for (int i = 1; i < argc - 1; ++i) {
         if (!strcmp(argv[i], "--port")) port = atoi(argv[++i]);
         else if (!strcmp(argv[i], "--data")) data_path = argv[++i];
         else if (!strcmp(argv[i], "--sigma")) w.sigma = atof(argv[++i]);
+        else if (!strcmp(argv[i], "--model")) logistic = !strcmp(argv[++i], "logistic");
         else if (!strcmp(argv[i], "--dtype")) {
             const char* d = argv[++i];
             w.dtype = !strcmp(d, "f64") ? FED_F64 : !strcmp(d, "f32") ? FED_F32 : FED_BF16;
@@ -356,15 +455,27 @@ int main(int argc, char** argv) {
         return 2;
     }
     w.eval = (eval_fn_t)dlsym(lib, "fed_gaussian_linear_eval");
+    w.eval_logistic = (logistic_fn_t)dlsym(lib, "fed_logistic_glm");
     host_alloc_fn_t host_alloc = (host_alloc_fn_t)dlsym(lib, "fed_host_alloc");
-    if (!w.eval || !host_alloc) {
+    if (!w.eval || !w.eval_logistic || !host_alloc) {
         fprintf(stderr, "missing symbols in %s\n", lib_path.c_str());
         return 2;
     }
-    if (!load_shard(w, data_path)) return 2;
-    if (hipMalloc(&w.out_dev, 3 * 8) != hipSuccess ||
+    if (logistic) {
+        if (!load_logistic_shard(w, data_path)) return 2;
+        w.out_host.resize(1 + w.K);
+        if (hipMalloc((void**)&w.out_dev, (1 + w.K) * 8) != hipSuccess ||
+            hipMalloc((void**)&w.beta_dev, w.K * 4) != hipSuccess ||
+            hipMalloc((void**)&w.ws_f32, (long long)1024 * w.K * 4) != hipSuccess) {
+            fprintf(stderr, "device alloc failed\n");
+            return 2;
+        }
+    } else if (!load_shard(w, data_path)) {
+        return 2;
+    }
+    if (!logistic && (hipMalloc(&w.out_dev, 3 * 8) != hipSuccess ||
         hipMalloc(&w.ws_dev, (72 + 3 * 2048) * 8) != hipSuccess ||
-        hipMemset(w.ws_dev, 0, (72 + 3 * 2048) * 8) != hipSuccess) {
+        hipMemset(w.ws_dev, 0, (72 + 3 * 2048) * 8) != hipSuccess)) {
         fprintf(stderr, "device alloc failed\n");
         return 2;
     }

@@ -77,3 +77,44 @@ def test_cpp_worker_serves_kernel_evals(tmp_path):
     finally:
         proc.terminate()
         proc.wait(timeout=10)
+
+
+@pytest.mark.timeout(300)
+def test_cpp_worker_serves_logistic(tmp_path):
+    from pytensor_federated_amd.common import LogpGradServiceClient
+    from pytensor_federated_amd.models import LogisticGLMModel, generate_logistic_dataset
+
+    import torch
+
+    X, y, beta0 = generate_logistic_dataset(20_000, 512, seed=52)
+    shard = tmp_path / "logit.bin"
+    with open(shard, "wb") as f:
+        f.write(struct.pack("<qq", X.shape[0], X.shape[1]))
+        f.write(np.asarray(X, dtype=np.float64).tobytes())
+        f.write(np.asarray(y, dtype=np.float64).tobytes())
+
+    env = dict(os.environ, FEDOPS_LIB=str(LIB))
+    proc = subprocess.Popen(
+        [str(WORKER), "--port", str(PORT + 2), "--data", str(shard),
+         "--model", "logistic", "--dtype", "bf16"],
+        env=env, stderr=subprocess.PIPE,
+    )
+    try:
+        _wait_tcp(PORT + 2)
+        client = LogpGradServiceClient("127.0.0.1", PORT + 2, transport="fast")
+        logp, (grad,) = client.evaluate(np.asarray(beta0, dtype=np.float64))
+        ref_model = LogisticGLMModel(
+            X, y, device="cuda:0", dtype=torch.bfloat16, use_kernels=True
+        )
+        # the worker quantizes beta to f32 inside the kernel wrapper; so does ours
+        logp_ref, (g_ref,) = ref_model.logp_grad(
+            torch.as_tensor(beta0, dtype=torch.float32)
+        )
+        np.testing.assert_allclose(float(logp), float(logp_ref), rtol=1e-6)
+        np.testing.assert_allclose(
+            grad, g_ref.cpu().numpy(), rtol=1e-5, atol=1e-4 * float(abs(g_ref).max())
+        )
+        del client
+    finally:
+        proc.terminate()
+        proc.wait(timeout=10)
