@@ -9,8 +9,10 @@ path is an explicit model-level opt-out, ``use_kernels=False``).
 from __future__ import annotations
 
 import ctypes
+import threading
 from typing import Optional, Tuple
 
+import numpy as _np
 import torch
 
 from .build import LIB_PATH
@@ -166,10 +168,6 @@ def gaussian_linear_logp_grad(
     _check(rc, "fed_gaussian_linear")
     return out[0], out[1], out[2]
 
-
-import threading
-
-import numpy as _np
 
 _ws_cache = {}
 _mailbox_lock = threading.Lock()
