@@ -156,7 +156,11 @@ def main():
         try:
             from pytensor_federated_amd.parallel.graphed import GraphedLinearEngine
 
-            graphed = GraphedLinearEngine(model, distributed=distributed)
+            # rank-safe: all ranks agree on the path (a one-sided capture
+            # failure would desynchronize the collective sequence)
+            graphed = GraphedLinearEngine.create_agreed(model, distributed=distributed)
+            if graphed is None and rank == 0:
+                print("# graphed path unavailable (agreed); engine fallback", flush=True)
         except Exception as ex:
             print(f"# graphed path unavailable ({ex}); falling back", flush=True)
             graphed = None

@@ -109,7 +109,46 @@ class GraphedLogpGradEngine:
 class GraphedLinearEngine:
     """Graph-replayed evaluator for a GaussianLinearModel shard."""
 
-    def __init__(self, model, distributed: bool = False, group=None) -> None:
+    @staticmethod
+    def create_agreed(model, distributed: bool = False, group=None):
+        """Rank-safe construction for multi-rank use.
+
+        A graph capture that fails on SOME ranks while others proceed would
+        desynchronize the collective sequence (the engine's warmup runs one
+        all-reduce) and deadlock the job.  Protocol: every rank first runs
+        one ALIGNED eager all-reduce (communicator init), then attempts the
+        capture, then all ranks agree (MIN-reduce of a success flag);
+        returns the engine only if EVERY rank captured, else None on all
+        ranks -- the caller falls back to the eager engine path uniformly.
+        """
+        if not distributed:
+            try:
+                return GraphedLinearEngine(model, distributed=False, group=group)
+            except Exception:
+                return None
+        import torch.distributed as dist
+
+        device = model._x.device
+        # aligned communicator warm-up (1 collective on every rank)
+        probe = torch.zeros(3, dtype=torch.float64, device=device)
+        dist.all_reduce(probe, group=group)
+        engine = None
+        try:
+            engine = GraphedLinearEngine(
+                model, distributed=True, group=group, skip_comm_warmup=True
+            )
+            ok = 1.0
+        except Exception:
+            engine = None
+            ok = 0.0
+        flag = torch.tensor([ok], dtype=torch.float64, device=device)
+        dist.all_reduce(flag, op=dist.ReduceOp.MIN, group=group)
+        if float(flag[0]) < 1.0:
+            return None
+        return engine
+
+    def __init__(self, model, distributed: bool = False, group=None,
+                 skip_comm_warmup: bool = False) -> None:
         from ..ops import (
             alloc_mailbox,
             gaussian_linear_launch_theta,
@@ -133,19 +172,21 @@ class GraphedLinearEngine:
         self.epoch_dev = torch.zeros(1, dtype=torch.int64, device=device)
         self._expected = 0
 
-        def body():
+        def body(include_comm: bool = True):
             self.theta_dev.copy_(self.theta_pinned, non_blocking=True)
             gaussian_linear_launch_theta(
                 model._x, model._y, self.theta_dev, model._sigma, self.buf, self.ws
             )
-            if self._distributed:
+            if self._distributed and include_comm:
                 import torch.distributed as dist
 
                 dist.all_reduce(self.buf, op=dist.ReduceOp.SUM, group=self._group)
             publish_result(self.buf, self.mailbox, self.epoch_dev)
 
-        # warmup (also establishes RCCL communicators before capture)
-        body()
+        # warmup; under create_agreed the communicator was already warmed by
+        # the ALIGNED probe, and this warmup must not issue collectives (a
+        # rank whose construction fails before them would desync the job)
+        body(include_comm=not skip_comm_warmup)
         torch.cuda.synchronize()
         self._expected = int(self._seq_view[0])
 
