@@ -129,6 +129,10 @@ def main():
     theta_dev = torch.as_tensor(theta0, dtype=torch.float64, device=device)
     host_buf = None
 
+    # path override for debugging/driver triage:
+    #   PFA_BENCH_PATH=persistent|graphed|sync-native|engine
+    path_override = os.environ.get("PFA_BENCH_PATH", "")
+
     # persistent eval-server path (single-GPU linear): resident kernel,
     # pinned request/result mailboxes -- no launch, no ramp per call
     persistent = None
@@ -138,6 +142,7 @@ def main():
         and not distributed
         and use_kernels is None
         and readback
+        and path_override in ("", "persistent")
     ):
         try:
             from pytensor_federated_amd.ops import PersistentLinearEngine
@@ -150,7 +155,7 @@ def main():
     # graphed paths: hipGraph replay of [H2D theta -> model eval ->
     # RCCL all-reduce -> mailbox publish] per evaluation
     graphed = None
-    if persistent is not None:
+    if persistent is not None or path_override in ("sync-native", "engine"):
         pass
     elif args.model == "linear" and have_gpu and use_kernels is None and readback:
         try:
@@ -176,10 +181,12 @@ def main():
     # single-GPU linear serving path: one native sync call per evaluation
     fast_sync = (
         graphed is None
+        and persistent is None
         and not distributed
         and readback
         and args.model == "linear"
         and hasattr(model, "logp_grad_sync")
+        and path_override != "engine"
     )
 
     def one_step(t: int):
