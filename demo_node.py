@@ -12,7 +12,8 @@ import argparse
 import multiprocessing
 
 
-def run_node(bind: str, port: int, delay: float, device: str, rows: int, seed: int):
+def run_node(bind: str, port: int, delay: float, device: str, rows: int, seed: int,
+             fast_port: int = None):
     import torch
 
     from pytensor_federated_amd.common import wrap_logp_grad_func
@@ -22,15 +23,27 @@ def run_node(bind: str, port: int, delay: float, device: str, rows: int, seed: i
     x, y = generate_linear_dataset(rows, seed=seed)
     dtype = torch.bfloat16 if device.startswith("cuda") else torch.float64
     model = GaussianLinearModel(x, y, sigma=0.4, device=device, dtype=dtype, delay=delay or None)
-    print(f"Serving linear-model shard ({rows} rows, seed {seed}) on {bind}:{port} [{device}]")
-    serve_compute_func(wrap_logp_grad_func(model.as_logp_grad_func()), bind, port)
+    print(f"Serving linear-model shard ({rows} rows, seed {seed}) on {bind}:{port} "
+          f"[{device}]" + (f" + fast:{fast_port}" if fast_port else ""))
+    serve_compute_func(
+        wrap_logp_grad_func(model.as_logp_grad_func()), bind, port, fast_port=fast_port
+    )
 
 
-def run_node_pool(bind: str, ports, delay: float, device: str, rows: int):
-    """One server process per port (reference demo_node.py:98-121)."""
+def run_node_pool(bind: str, ports, delay: float, device: str, rows: int,
+                  fast_offset: int = 0):
+    """One server process per port (reference demo_node.py:98-121).
+
+    ``fast_offset`` > 0 additionally serves the low-latency fast transport
+    on port+offset for each worker."""
     ctx = multiprocessing.get_context("spawn")
     procs = [
-        ctx.Process(target=run_node, args=(bind, port, delay, device, rows, i), daemon=False)
+        ctx.Process(
+            target=run_node,
+            args=(bind, port, delay, device, rows, i,
+                  port + fast_offset if fast_offset else None),
+            daemon=False,
+        )
         for i, port in enumerate(ports)
     ]
     for p in procs:
@@ -50,5 +63,10 @@ if __name__ == "__main__":
     parser.add_argument("--delay", type=float, default=0.0)
     parser.add_argument("--device", default="cpu", help='"cpu" or "cuda:0"')
     parser.add_argument("--rows", type=int, default=10)
+    parser.add_argument(
+        "--fast-offset", type=int, default=0,
+        help="also serve the fast transport on port+offset (0 = off)",
+    )
     args, _ = parser.parse_known_args()
-    run_node_pool(args.bind, args.ports, args.delay, args.device, args.rows)
+    run_node_pool(args.bind, args.ports, args.delay, args.device, args.rows,
+                  args.fast_offset)

@@ -50,6 +50,7 @@ from .service import (  # noqa: F401
 from .signatures import ComputeFunc, LogpFunc, LogpGradFunc  # noqa: F401
 
 try:
+    from . import inference, models, parallel  # noqa: F401
     from .torch_ops import LogpGradOp, LogpOp, federated_logp_grad  # noqa: F401
 except ModuleNotFoundError:
     # torch not installed: transport-only deployment (e.g. a CPU client box).

@@ -392,3 +392,13 @@ def test_persistent_engine_relaunch_after_idle(dev):
             assert eng.logp_grad_sync(1.2, 0.4) == ref
     finally:
         eng.close()
+
+
+def test_get_load_reports_gpu_telemetry(dev):
+    from pytensor_federated_amd.service import ArraysToArraysService
+
+    svc = ArraysToArraysService(lambda a: [a], report_gpu_load=True)
+    load = svc.determine_load()
+    # amdsmi-backed: HBM% in [0,100]; busy% may be 0 on an idle box
+    assert 0.0 <= load.percent_ram <= 100.0
+    assert 0.0 <= load.percent_cpu <= 100.0
