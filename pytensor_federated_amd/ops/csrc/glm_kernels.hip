@@ -333,7 +333,7 @@ __global__ __launch_bounds__(256) void k_gaussian_linear_fused(
 // block writes one fp32 slab row: slab[block][K].  k_colsum_reduce sums the
 // slab into out[1..K] (deterministic; no fp32 global atomics).
 
-template <typename T, int KITER, int NROWS>  // KITER = K/(WAVE*VEC); NROWS rows in flight
+template <typename T, int KITER>  // KITER = K / (WAVE*VEC), compile-time
 __global__ __launch_bounds__(256) void k_logistic_glm_reg(
     const T* __restrict__ X,   // [N][K] row-major
     const T* __restrict__ y,
@@ -351,8 +351,10 @@ __global__ __launch_bounds__(256) void k_logistic_glm_reg(
     const long long wave_id = (long long)blockIdx.x * waves_per_block + wid;
     const long long n_waves = (long long)gridDim.x * waves_per_block;
 
-    float breg[KITER][VEC];
-    float gacc[KITER][VEC];
+    // per-lane register state
+    float xreg[KITER][VEC];        // this lane's row slice
+    float breg[KITER][VEC];        // this lane's beta slice (loop-invariant)
+    float gacc[KITER][VEC];        // this lane's grad accumulator
 #pragma unroll
     for (int c = 0; c < KITER; ++c)
 #pragma unroll
@@ -362,65 +364,61 @@ __global__ __launch_bounds__(256) void k_logistic_glm_reg(
         }
 
     double logp_acc = 0.0;
-    // NROWS rows per iteration: all rows' global loads issue before any
-    // dependent dot/reduce/transcendental chain (1-row loop measured 68%
-    // of HBM peak; 2 rows 82%).
-    float xr[NROWS][KITER][VEC];
-    const long long stride = n_waves * NROWS;
-    long long r = wave_id * NROWS;
-    for (; r + NROWS - 1 < n_rows; r += stride) {
+    // Two rows per iteration: both rows' global loads issue before either
+    // row's dependent dot/reduce/transcendental chain, so each wave keeps
+    // ~2x the HBM traffic in flight (the 1-row loop measured 68% of HBM
+    // peak; the dependent per-row chain was the gap).
+    float xreg2[KITER][VEC];
+    const long long pair_stride = n_waves * 2;
+    long long r = wave_id * 2;
+    for (; r + 1 < n_rows; r += pair_stride) {
+        const T* row0 = X + r * (long long)K;
+        const T* row1 = row0 + K;
+        float z0p = 0.f, z1p = 0.f;
 #pragma unroll
-        for (int q = 0; q < NROWS; ++q) {
-            const T* row = X + (r + q) * (long long)K;
-#pragma unroll
-            for (int c = 0; c < KITER; ++c) {
-                // nontemporal: each X row is consumed once (registers carry
-                // it between the dot and the grad update)
-                const u32x4_t v = __builtin_nontemporal_load(
-                    (const u32x4_t*)(row + c * WAVE * VEC + lane * VEC));
-                TR::unpack(v, xr[q][c]);
-            }
-        }
-        float zp[NROWS];
-#pragma unroll
-        for (int q = 0; q < NROWS; ++q) zp[q] = 0.f;
-#pragma unroll
-        for (int c = 0; c < KITER; ++c)
-#pragma unroll
-            for (int j = 0; j < VEC; ++j)
-#pragma unroll
-                for (int q = 0; q < NROWS; ++q) zp[q] += xr[q][c][j] * breg[c][j];
-        float z[NROWS], res[NROWS];
-#pragma unroll
-        for (int q = 0; q < NROWS; ++q) {
-            z[q] = wave_reduce_sum_f32(zp[q]);
-            z[q] = __shfl(z[q], 0, WAVE);
-        }
-#pragma unroll
-        for (int q = 0; q < NROWS; ++q) {
-            const float yr = TR::get(y, r + q);
-            const float sp = fmaxf(z[q], 0.f) + log1pf(__expf(-fabsf(z[q])));
-            if (lane == 0) logp_acc += (double)(yr * z[q] - sp);
-            res[q] = yr - 1.f / (1.f + __expf(-z[q]));
+        for (int c = 0; c < KITER; ++c) {
+            // nontemporal: each X row is consumed once (from registers for
+            // both logp and grad) -- do not displace L2/L3 lines
+            const u32x4_t v0 = __builtin_nontemporal_load(
+                (const u32x4_t*)(row0 + c * WAVE * VEC + lane * VEC));
+            const u32x4_t v1 = __builtin_nontemporal_load(
+                (const u32x4_t*)(row1 + c * WAVE * VEC + lane * VEC));
+            TR::unpack(v0, xreg[c]);
+            TR::unpack(v1, xreg2[c]);
         }
 #pragma unroll
         for (int c = 0; c < KITER; ++c)
 #pragma unroll
             for (int j = 0; j < VEC; ++j) {
-                float g = gacc[c][j];
-#pragma unroll
-                for (int q = 0; q < NROWS; ++q) g += res[q] * xr[q][c][j];
-                gacc[c][j] = g;
+                z0p += xreg[c][j] * breg[c][j];
+                z1p += xreg2[c][j] * breg[c][j];
             }
+        float z0 = wave_reduce_sum_f32(z0p);
+        float z1 = wave_reduce_sum_f32(z1p);
+        z0 = __shfl(z0, 0, WAVE);
+        z1 = __shfl(z1, 0, WAVE);
+        const float y0 = TR::get(y, r);
+        const float y1 = TR::get(y, r + 1);
+        // stable: y*z - softplus(z) = y*z - (max(z,0) + log1p(exp(-|z|)))
+        const float sp0 = fmaxf(z0, 0.f) + log1pf(__expf(-fabsf(z0)));
+        const float sp1 = fmaxf(z1, 0.f) + log1pf(__expf(-fabsf(z1)));
+        if (lane == 0) logp_acc += (double)(y0 * z0 - sp0) + (double)(y1 * z1 - sp1);
+        const float res0 = y0 - 1.f / (1.f + __expf(-z0));
+        const float res1 = y1 - 1.f / (1.f + __expf(-z1));
+#pragma unroll
+        for (int c = 0; c < KITER; ++c)
+#pragma unroll
+            for (int j = 0; j < VEC; ++j)
+                gacc[c][j] += res0 * xreg[c][j] + res1 * xreg2[c][j];
     }
-    for (; r < n_rows; ++r) {  // up to NROWS-1 tail rows of this wave
+    for (; r < n_rows; r += pair_stride) {  // odd tail row of this wave
         const T* row = X + r * (long long)K;
         float z_part = 0.f;
 #pragma unroll
         for (int c = 0; c < KITER; ++c) {
-            TR::load(row + c * WAVE * VEC + lane * VEC, xr[0][c]);
+            TR::load(row + c * WAVE * VEC + lane * VEC, xreg[c]);
 #pragma unroll
-            for (int j = 0; j < VEC; ++j) z_part += xr[0][c][j] * breg[c][j];
+            for (int j = 0; j < VEC; ++j) z_part += xreg[c][j] * breg[c][j];
         }
         float z = wave_reduce_sum_f32(z_part);
         z = __shfl(z, 0, WAVE);
@@ -431,7 +429,7 @@ __global__ __launch_bounds__(256) void k_logistic_glm_reg(
 #pragma unroll
         for (int c = 0; c < KITER; ++c)
 #pragma unroll
-            for (int j = 0; j < VEC; ++j) gacc[c][j] += resid * xr[0][c][j];
+            for (int j = 0; j < VEC; ++j) gacc[c][j] += resid * xreg[c][j];
     }
 
     // cross-wave grad reduction in LDS (dynamic: K floats), then one slab row
@@ -634,23 +632,10 @@ int fed_logistic_glm(
     if (grid < 1) return -3;
     const int lds_bytes = ((K * 4 + 15) & ~15) + waves_per_block * 8;
 
-    static int nrows_sel = 0;
-    if (nrows_sel == 0) {
-        const char* env = getenv("FED_LOGISTIC_ROWS");
-        nrows_sel = env ? atoi(env) : 2;
-        if (nrows_sel != 3) nrows_sel = 2;
-    }
 #define LAUNCH_LOGISTIC(T, KITER)                                                      \
-    do {                                                                               \
-        if (nrows_sel == 3)                                                            \
-            hipLaunchKernelGGL((k_logistic_glm_reg<T, KITER, 3>), dim3(grid),          \
-                               dim3(block), lds_bytes, stream, (const T*)X,            \
-                               (const T*)y, n_rows, K, beta_f32, out, workspace);      \
-        else                                                                           \
-            hipLaunchKernelGGL((k_logistic_glm_reg<T, KITER, 2>), dim3(grid),          \
-                               dim3(block), lds_bytes, stream, (const T*)X,            \
-                               (const T*)y, n_rows, K, beta_f32, out, workspace);      \
-    } while (0)
+    hipLaunchKernelGGL((k_logistic_glm_reg<T, KITER>), dim3(grid), dim3(block),        \
+                       lds_bytes, stream, (const T*)X, (const T*)y, n_rows, K,         \
+                       beta_f32, out, workspace)
 
     if (dtype == FED_BF16) {
         switch (K) {
