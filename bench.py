@@ -173,7 +173,9 @@ def main():
         try:
             from pytensor_federated_amd.parallel.graphed import GraphedLogpGradEngine
 
-            graphed = GraphedLogpGradEngine(model, (4,), distributed=distributed)
+            graphed = GraphedLogpGradEngine.create_agreed(model, (4,), distributed=distributed)
+            if graphed is None and rank == 0:
+                print("# graphed ODE path unavailable (agreed); engine fallback", flush=True)
         except Exception as ex:
             print(f"# graphed ODE path unavailable ({ex}); falling back", flush=True)
             graphed = None

@@ -34,7 +34,33 @@ class GraphedLogpGradEngine:
     ``logp_grad`` is shape-static and sync-free (ODEModel qualifies).
     """
 
-    def __init__(self, model, theta_shape, distributed: bool = False, group=None) -> None:
+    @staticmethod
+    def create_agreed(model, theta_shape, distributed: bool = False, group=None):
+        """Rank-safe construction (see GraphedLinearEngine.create_agreed)."""
+        if not distributed:
+            try:
+                return GraphedLogpGradEngine(model, theta_shape, distributed=False, group=group)
+            except Exception:
+                return None
+        import torch.distributed as dist
+
+        device = model.device
+        probe = torch.zeros(1, dtype=torch.float64, device=device)
+        dist.all_reduce(probe, group=group)  # aligned communicator warm-up
+        engine = None
+        try:
+            engine = GraphedLogpGradEngine(
+                model, theta_shape, distributed=True, group=group, skip_comm_warmup=True
+            )
+            ok = 1.0
+        except Exception:
+            ok = 0.0
+        flag = torch.tensor([ok], dtype=torch.float64, device=device)
+        dist.all_reduce(flag, op=dist.ReduceOp.MIN, group=group)
+        return engine if float(flag[0]) >= 1.0 else None
+
+    def __init__(self, model, theta_shape, distributed: bool = False, group=None,
+                 skip_comm_warmup: bool = False) -> None:
         from ..ops import alloc_mailbox, publish_result
 
         self.model = model
@@ -55,7 +81,7 @@ class GraphedLogpGradEngine:
         self.epoch_dev = torch.zeros(1, dtype=torch.int64, device=device)
         self._n_out = n_out
 
-        def body():
+        def body(include_comm: bool = True):
             self.theta_dev.copy_(
                 self.theta_pinned.reshape(self.theta_dev.shape), non_blocking=True
             )
@@ -66,13 +92,15 @@ class GraphedLogpGradEngine:
                 n = g.numel()
                 self.buf[off : off + n] = g.reshape(-1).to(torch.float64)
                 off += n
-            if self._distributed:
+            if self._distributed and include_comm:
                 import torch.distributed as dist
 
                 dist.all_reduce(self.buf, op=dist.ReduceOp.SUM, group=self._group)
             publish_result(self.buf, self.mailbox, self.epoch_dev)
 
-        body()  # warmup (cuBLAS/RCCL init etc.)
+        # warmup; no collectives here under create_agreed (a rank whose
+        # construction fails before them would desync the job)
+        body(include_comm=not skip_comm_warmup)
         torch.cuda.synchronize()
         self._expected = int(self._seq_view[0])
         self.graph = torch.cuda.CUDAGraph()
