@@ -68,8 +68,12 @@ async def start_fast_server_async(service, bind: str, port: int):
                 ftype, payload = await _read_frame(reader)
                 if ftype == T_EVAL:
                     try:
+                        exporter = getattr(service, "_exporter", None)
+                        if exporter is not None:
+                            exporter.reset()
                         out = _run_compute_func(
-                            InputArrays.FromString(payload), service._compute_func
+                            InputArrays.FromString(payload), service._compute_func,
+                            exporter,
                         )
                         writer.write(_frame(T_EVAL_R, out.SerializeToString()))
                     except Exception as ex:  # surface compute errors to client

@@ -63,18 +63,42 @@ __all__ = [
 _BALANCE_DESYNC_RANGE = (0.2, 2.0)
 
 
-def _run_compute_func(input_arrays: InputArrays, compute_func: ComputeFunc) -> OutputArrays:
+def _run_compute_func(
+    input_arrays: InputArrays, compute_func: ComputeFunc, exporter=None
+) -> OutputArrays:
     """Decode inputs, run the compute function, encode outputs, echo the uuid.
 
     Parity: reference service.py:45-72.  Decoding is a zero-copy numpy view
     over the message bytes; encoding copies (the protobuf owns its bytes).
+
+    Device arrays: items with dtype ``hipipc/...`` (npproto.device) decode
+    to CUDA torch tensors via the dmabuf IPC handle -- HBM -> HBM, no host
+    staging.  With an ``exporter`` set, torch CUDA outputs are encoded the
+    same way (on-node replies never round-trip through host bytes).
     """
-    inputs = [ndarray_to_numpy(item) for item in input_arrays.items]
+    inputs = []
+    for item in input_arrays.items:
+        if item.dtype.startswith("hipipc/"):
+            from .npproto.device import device_ndarray_to_torch
+
+            inputs.append(device_ndarray_to_torch(item))
+        else:
+            inputs.append(ndarray_to_numpy(item))
     outputs = compute_func(*inputs)
-    return OutputArrays(
-        items=[ndarray_from_numpy(np.asarray(o)) for o in outputs],
-        uuid=input_arrays.uuid,
-    )
+    items = []
+    for o in outputs:
+        is_cuda_tensor = (
+            type(o).__module__ == "torch" and getattr(o, "is_cuda", False)
+        )
+        if is_cuda_tensor and exporter is not None:
+            items.append(exporter.export(o))
+        elif is_cuda_tensor:
+            from .npproto.utils import ndarray_from_torch
+
+            items.append(ndarray_from_torch(o))
+        else:
+            items.append(ndarray_from_numpy(np.asarray(o)))
+    return OutputArrays(items=items, uuid=input_arrays.uuid)
 
 
 class ArraysToArraysService:
@@ -86,12 +110,26 @@ class ArraysToArraysService:
     the message stays wire-compatible.
     """
 
-    def __init__(self, compute_func: ComputeFunc, *, report_gpu_load: bool = False) -> None:
+    def __init__(
+        self,
+        compute_func: ComputeFunc,
+        *,
+        report_gpu_load: bool = False,
+        device_arrays: bool = False,
+    ) -> None:
+        """``device_arrays=True``: reply tensors that live on a ROCm device
+        are shipped as dmabuf IPC handles (HBM->HBM for same-node clients)
+        instead of host byte copies."""
         import psutil
 
         self._compute_func = compute_func
         self._n_clients = 0
         self._report_gpu_load = report_gpu_load
+        self._exporter = None
+        if device_arrays:
+            from .npproto.device import DeviceArrayExporter
+
+            self._exporter = DeviceArrayExporter()
         # Prime psutil's CPU monitoring so the first GetLoad is meaningful.
         psutil.getloadavg()
 
@@ -118,14 +156,18 @@ class ArraysToArraysService:
 
     # -- RPC handlers (grpc.aio behavior functions) --------------------
     async def evaluate(self, input_arrays: InputArrays, context=None) -> OutputArrays:
-        return _run_compute_func(input_arrays, self._compute_func)
+        if self._exporter is not None:
+            self._exporter.reset()  # previous reply was consumed (1 in flight)
+        return _run_compute_func(input_arrays, self._compute_func, self._exporter)
 
     async def evaluate_stream(self, request_iterator, context=None):
         self._n_clients += 1
         _log.info("A client started a stream. Now serving %i clients.", self._n_clients)
         try:
             async for input_arrays in request_iterator:
-                yield _run_compute_func(input_arrays, self._compute_func)
+                if self._exporter is not None:
+                    self._exporter.reset()
+                yield _run_compute_func(input_arrays, self._compute_func, self._exporter)
         finally:
             self._n_clients -= 1
             _log.info("A client ended a stream. Now serving %i clients.", self._n_clients)
@@ -441,6 +483,7 @@ class ArraysToArraysServiceClient:
         use_stream: bool = True,
         retries: int = 2,
         transport: str = "grpc",
+        device_arrays: bool = False,
     ) -> None:
         """
         Parameters
@@ -462,6 +505,10 @@ class ArraysToArraysServiceClient:
             or "fast" (this framework's raw-asyncio framing -- same protobuf
             payloads, ~10-20x lower per-call latency than grpc's C-core in
             containerized deployments).
+        device_arrays : bool
+            Ship torch CUDA tensor inputs as dmabuf IPC handles (HBM->HBM,
+            same-node worker) instead of host byte copies, and return CUDA
+            tensors for device-array replies.
         """
         if hosts_and_ports is None and (host is None or port is None):
             raise ValueError("Provide either host+port or hosts_and_ports.")
@@ -471,6 +518,8 @@ class ArraysToArraysServiceClient:
         self._use_stream = use_stream
         self._retries = retries
         self._transport = transport
+        self._device_arrays = device_arrays
+        self._exporter = None
 
     def __del__(self):
         cid = thread_pid_id(self)
@@ -524,16 +573,41 @@ class ArraysToArraysServiceClient:
         if retries is None:
             retries = self._retries
 
-        input_arrays = InputArrays(
-            items=[ndarray_from_numpy(np.asarray(i)) for i in inputs],
-            uuid=str(uuid_module.uuid4()),
-        )
+        items = []
+        for i in inputs:
+            if (
+                self._device_arrays
+                and type(i).__module__ == "torch"
+                and getattr(i, "is_cuda", False)
+            ):
+                if self._exporter is None:
+                    from .npproto.device import DeviceArrayExporter
+
+                    self._exporter = DeviceArrayExporter()
+                items.append(self._exporter.export(i))
+            else:
+                items.append(ndarray_from_numpy(np.asarray(i)))
+        if self._exporter is not None and not any(
+            it.dtype.startswith("hipipc/") for it in items
+        ):
+            pass  # nothing exported this call
+        input_arrays = InputArrays(items=items, uuid=str(uuid_module.uuid4()))
         cid = thread_pid_id(self)
         last_error: Optional[BaseException] = None
         for attempt in range(retries + 1):
             try:
                 output = await _connect_evaluate_async(self, input_arrays, use_stream)
-                return [ndarray_to_numpy(item) for item in output.items]
+                decoded = []
+                for item in output.items:
+                    if item.dtype.startswith("hipipc/"):
+                        from .npproto.device import device_ndarray_to_torch
+
+                        decoded.append(device_ndarray_to_torch(item))
+                    else:
+                        decoded.append(ndarray_to_numpy(item))
+                if self._exporter is not None:
+                    self._exporter.reset()  # server consumed the request arrays
+                return decoded
             except (grpc.RpcError, ConnectionError, OSError) as ex:
                 last_error = ex
                 _log.warning(

@@ -62,3 +62,67 @@ def test_device_export_metadata():
     nda2 = exporter.export(src)
     assert nda2.data[:64] == nda.data[:64]
     assert nda2.data != nda.data
+
+
+def _device_worker(port, result_q):
+    """Worker serving a GPU matmul over the fast transport with device-array
+    replies (HBM->HBM both directions)."""
+    try:
+        import asyncio
+
+        import torch
+
+        from pytensor_federated_amd.fastsock import start_fast_server_async
+        from pytensor_federated_amd.service import ArraysToArraysService
+
+        def matmul(A, B):
+            # inputs arrive as CUDA tensors (device codec); stay on device
+            assert isinstance(A, torch.Tensor) and A.is_cuda, type(A)
+            return [A.float() @ B.float()]
+
+        async def main():
+            service = ArraysToArraysService(matmul, device_arrays=True)
+            server = await start_fast_server_async(service, "127.0.0.1", port)
+            result_q.put("up")
+            async with server:
+                await server.serve_forever()
+
+        asyncio.run(main())
+    except Exception as ex:
+        result_q.put(f"err: {ex!r}")
+
+
+@pytest.mark.timeout(300)
+def test_device_arrays_through_transport():
+    """Full on-node device transport: client CUDA tensors -> IPC handles ->
+    worker computes on device -> IPC reply -> client CUDA tensor."""
+    import torch
+
+    from pytensor_federated_amd.service import ArraysToArraysServiceClient
+
+    ctx = multiprocessing.get_context("spawn")
+    q = ctx.Queue()
+    port = 9671
+    proc = ctx.Process(target=_device_worker, args=(port, q), daemon=True)
+    proc.start()
+    try:
+        assert q.get(timeout=240) == "up"
+        client = ArraysToArraysServiceClient(
+            "127.0.0.1", port, transport="fast", device_arrays=True
+        )
+        A = torch.randn(256, 512, device="cuda:0")
+        B = torch.randn(512, 128, device="cuda:0")
+        (C,) = client.evaluate(A, B)
+        assert isinstance(C, torch.Tensor) and C.is_cuda
+        torch.cuda.synchronize()
+        ref = A @ B
+        assert torch.allclose(C, ref, rtol=1e-4, atol=1e-4)
+        # repeated calls reuse the IPC mappings
+        for _ in range(10):
+            (C,) = client.evaluate(A, B)
+        torch.cuda.synchronize()
+        assert torch.allclose(C, ref, rtol=1e-4, atol=1e-4)
+        del client
+    finally:
+        proc.terminate()
+        proc.join(timeout=10)
