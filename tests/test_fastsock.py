@@ -145,3 +145,44 @@ def test_fast_latency_beats_grpc_floor(fast_servers):
     per_call = asyncio.run(run())
     assert per_call < 0.002, f"fast transport too slow: {per_call * 1e6:.0f} us/call"
     del client
+
+
+@pytest.mark.timeout(120)
+def test_fast_server_survives_malformed_frames(fast_servers):
+    """Garbage frames must not kill the worker: bad magic drops the
+    connection; unknown frame types get T_ERR; the server keeps serving."""
+    import asyncio as aio
+
+    from pytensor_federated_amd.fastsock import MAGIC, _frame
+
+    async def run():
+        # 1. wrong magic -> connection closed, server alive
+        r, w = await aio.open_connection("127.0.0.1", FAST_PORTS[0])
+        w.write(b"BOGUS")
+        await w.drain()
+        data = await r.read(64)
+        assert data == b""  # closed
+        w.close()
+
+        # 2. unknown frame type -> T_ERR reply
+        r, w = await aio.open_connection("127.0.0.1", FAST_PORTS[0])
+        w.write(MAGIC + _frame(0x7C, b"???"))
+        await w.drain()
+        hdr = await r.readexactly(5)
+        assert hdr[0] == 0xFF
+        w.close()
+
+        # 3. truncated protobuf payload -> T_ERR (compute error), not a crash
+        r, w = await aio.open_connection("127.0.0.1", FAST_PORTS[0])
+        w.write(MAGIC + _frame(0x01, b"\xff\xff\xff"))
+        await w.drain()
+        hdr = await r.readexactly(5)
+        assert hdr[0] == 0xFF
+        w.close()
+
+    asyncio.run(run())
+    # server still healthy
+    client = ArraysToArraysServiceClient("127.0.0.1", FAST_PORTS[0], transport="fast")
+    (out,) = client.evaluate(np.array(2.0), np.array(3.0))
+    np.testing.assert_array_equal(out, np.array(6.0))
+    del client
