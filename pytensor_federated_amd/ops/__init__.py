@@ -96,6 +96,12 @@ def _try_load() -> Optional[ctypes.CDLL]:
         ctypes.c_int, ctypes.c_int, ctypes.c_double, ctypes.c_double,
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
     ]
+    lib.fed_ode_lv_eval_batched.restype = ctypes.c_int
+    lib.fed_ode_lv_eval_batched.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_double, ctypes.c_double,
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+    ]
     lib.fed_logistic_glm_batched.restype = ctypes.c_int
     lib.fed_logistic_glm_batched.argtypes = [
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_longlong, ctypes.c_int,
@@ -334,6 +340,33 @@ class PersistentLinearEngine:
             self.close()
         except Exception:
             pass
+
+
+def ode_lv_logp_grad_batched(
+    u0: torch.Tensor,
+    y_obs: torch.Tensor,
+    obs_of_step: torch.Tensor,
+    n_steps: int,
+    h: float,
+    sigma: float,
+    theta_c: torch.Tensor,     # [C, 4] f64
+    states_ws: torch.Tensor,   # [C*(n_steps+1)*B*2] f64 scratch
+    out: Optional[torch.Tensor] = None,  # f64[C*5]
+) -> torch.Tensor:
+    """Batched native Lotka-Volterra: C thetas in one sweep -> [C][5]."""
+    lib = require_kernels()
+    B = u0.shape[0]
+    C = theta_c.shape[0]
+    theta_dev = theta_c.detach().to(device=u0.device, dtype=torch.float64).contiguous()
+    if out is None:
+        out = torch.empty(C * 5, dtype=torch.float64, device=u0.device)
+    rc = lib.fed_ode_lv_eval_batched(
+        u0.data_ptr(), y_obs.data_ptr(), obs_of_step.data_ptr(),
+        int(n_steps), int(B), int(C), float(h), float(sigma),
+        theta_dev.data_ptr(), states_ws.data_ptr(), out.data_ptr(), _stream_ptr(),
+    )
+    _check(rc, "fed_ode_lv_eval_batched")
+    return out.reshape(C, 5)
 
 
 BATCH_CHAINS = 16

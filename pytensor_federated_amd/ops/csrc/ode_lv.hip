@@ -193,6 +193,153 @@ __global__ __launch_bounds__(256) void k_lv_adjoint(
     }
 }
 
+// Batched variant: C chains (thetas) x B experiments in one sweep.  Lane
+// l handles (chain = l / B, experiment = l % B); with B a multiple of 256
+// every block belongs to one chain, so the block reduction can target that
+// chain's output slot directly.  Fills the chip (B*C lanes) where the
+// single-theta kernel at B=1024 only occupies 4 CUs.
+__global__ __launch_bounds__(256) void k_lv_forward_batched(
+    const double* __restrict__ u0,       // [B][2]
+    const double* __restrict__ y_obs,    // [n_obs][B][2]
+    const int* __restrict__ obs_of_step,
+    int n_steps, int B, int n_chains,
+    double h, double inv_sig2,
+    const double* __restrict__ theta,    // [C][4]
+    double* __restrict__ states,         // [C][n_steps+1][B][2]
+    double* __restrict__ out             // [C][5] pre-zeroed
+) {
+    const long long total = (long long)B * n_chains;
+    for (long long l = blockIdx.x * blockDim.x + threadIdx.x; l < total;
+         l += (long long)gridDim.x * blockDim.x) {
+        const int ch = (int)(l / B);
+        const int e = (int)(l % B);
+        const double* th = theta + 4 * ch;
+        double thr[4] = {th[0], th[1], th[2], th[3]};
+        double* st_base = states + (size_t)ch * (n_steps + 1) * B * 2;
+        double x = u0[2 * e], y = u0[2 * e + 1];
+        st_base[2 * e] = x;
+        st_base[2 * e + 1] = y;
+        double logp_acc = 0.0;
+        if (obs_of_step[0] >= 0) {
+            const double* yo = y_obs + (size_t)obs_of_step[0] * B * 2 + 2 * e;
+            const double rx = yo[0] - x, ry = yo[1] - y;
+            logp_acc -= 0.5 * inv_sig2 * (rx * rx + ry * ry);
+        }
+        for (int s = 1; s <= n_steps; ++s) {
+            Rk4Mid m;
+            double nx, ny;
+            rk4_fwd(x, y, h, thr, m, nx, ny);
+            x = nx;
+            y = ny;
+            double* st = st_base + (size_t)s * B * 2 + 2 * e;
+            st[0] = x;
+            st[1] = y;
+            const int j = obs_of_step[s];
+            if (j >= 0) {
+                const double* yo = y_obs + (size_t)j * B * 2 + 2 * e;
+                const double rx = yo[0] - x, ry = yo[1] - y;
+                logp_acc -= 0.5 * inv_sig2 * (rx * rx + ry * ry);
+            }
+        }
+        // per-lane atomic into the chain slot (few k lanes per chain; the
+        // f64 atomic rate is ample at this arithmetic intensity)
+        atomicAdd(&out[5 * ch], logp_acc);
+    }
+}
+
+__global__ __launch_bounds__(256) void k_lv_adjoint_batched(
+    const double* __restrict__ y_obs,
+    const int* __restrict__ obs_of_step,
+    int n_steps, int B, int n_chains,
+    double h, double inv_sig2,
+    const double* __restrict__ theta,
+    const double* __restrict__ states,
+    double* __restrict__ out  // [C][5]
+) {
+    const long long total = (long long)B * n_chains;
+    for (long long l = blockIdx.x * blockDim.x + threadIdx.x; l < total;
+         l += (long long)gridDim.x * blockDim.x) {
+        const int ch = (int)(l / B);
+        const int e = (int)(l % B);
+        const double* th = theta + 4 * ch;
+        double thr[4] = {th[0], th[1], th[2], th[3]};
+        const double* st_base = states + (size_t)ch * (n_steps + 1) * B * 2;
+        double gth[4] = {0, 0, 0, 0};
+        double lx = 0.0, ly = 0.0;
+        {
+            const int j = obs_of_step[n_steps];
+            if (j >= 0) {
+                const double* st = st_base + (size_t)n_steps * B * 2 + 2 * e;
+                const double* yo = y_obs + (size_t)j * B * 2 + 2 * e;
+                lx += inv_sig2 * (yo[0] - st[0]);
+                ly += inv_sig2 * (yo[1] - st[1]);
+            }
+        }
+        for (int s = n_steps - 1; s >= 0; --s) {
+            const double* st = st_base + (size_t)s * B * 2 + 2 * e;
+            const double x = st[0], y = st[1];
+            Rk4Mid m;
+            double nx, ny;
+            rk4_fwd(x, y, h, thr, m, nx, ny);
+            const double w1 = h / 6.0, w24 = h / 3.0;
+            double gx = 0, gy = 0;
+            double g4x = 0, g4y = 0;
+            lv_vjp(m.u4x, m.u4y, thr, w1 * lx, w1 * ly, g4x, g4y, gth);
+            gx += g4x;
+            gy += g4y;
+            double c3x = w24 * lx + h * g4x, c3y = w24 * ly + h * g4y;
+            double g3x = 0, g3y = 0;
+            lv_vjp(m.u3x, m.u3y, thr, c3x, c3y, g3x, g3y, gth);
+            gx += g3x;
+            gy += g3y;
+            double c2x = w24 * lx + 0.5 * h * g3x, c2y = w24 * ly + 0.5 * h * g3y;
+            double g2x = 0, g2y = 0;
+            lv_vjp(m.u2x, m.u2y, thr, c2x, c2y, g2x, g2y, gth);
+            gx += g2x;
+            gy += g2y;
+            double c1x = w1 * lx + 0.5 * h * g2x, c1y = w1 * ly + 0.5 * h * g2y;
+            double g1x = 0, g1y = 0;
+            lv_vjp(x, y, thr, c1x, c1y, g1x, g1y, gth);
+            gx += g1x;
+            gy += g1y;
+            lx += gx;
+            ly += gy;
+            const int j = obs_of_step[s];
+            if (j >= 0) {
+                const double* yo = y_obs + (size_t)j * B * 2 + 2 * e;
+                lx += inv_sig2 * (yo[0] - x);
+                ly += inv_sig2 * (yo[1] - y);
+            }
+        }
+#pragma unroll
+        for (int k = 0; k < 4; ++k) atomicAdd(&out[5 * ch + 1 + k], gth[k]);
+    }
+}
+
+extern "C" int fed_ode_lv_eval_batched(
+    const double* u0, const double* y_obs, const int* obs_of_step,
+    int n_steps, int B, int n_chains, double h, double sigma,
+    const double* theta_dev,  // [C][4]
+    double* states_ws, double* out, void* stream_v
+) {
+    hipStream_t stream = (hipStream_t)stream_v;
+    hipError_t err = hipMemsetAsync(out, 0, (size_t)n_chains * 5 * sizeof(double), stream);
+    if (err != hipSuccess) return (int)err;
+    const double inv_sig2 = 1.0 / (sigma * sigma);
+    long long total = (long long)B * n_chains;
+    int grid = (int)((total + 255) / 256);
+    if (grid > 2048) grid = 2048;
+    hipLaunchKernelGGL(k_lv_forward_batched, dim3(grid), dim3(256), 0, stream,
+                       u0, y_obs, obs_of_step, n_steps, B, n_chains, h, inv_sig2,
+                       theta_dev, states_ws, out);
+    hipError_t e1 = hipGetLastError();
+    if (e1 != hipSuccess) return (int)e1;
+    hipLaunchKernelGGL(k_lv_adjoint_batched, dim3(grid), dim3(256), 0, stream,
+                       y_obs, obs_of_step, n_steps, B, n_chains, h, inv_sig2,
+                       theta_dev, states_ws, out);
+    return (int)hipGetLastError();
+}
+
 extern "C" int fed_ode_lv_eval(
     const double* u0, const double* y_obs, const int* obs_of_step,
     int n_steps, int B, double h, double sigma,

@@ -422,3 +422,20 @@ def test_mala_16_chains_on_batched_kernel(dev):
     post_mean = chain[50:].mean(axis=(0, 2))
     # N=2e5 rows: posterior concentrates near the truth
     assert np.corrcoef(post_mean, beta_true)[0, 1] > 0.9
+
+
+def test_ode_batched_chains_native(dev):
+    from pytensor_federated_amd.models import ODEModel
+    from pytensor_federated_amd.models.ode import generate_ode_dataset, lotka_volterra_rhs
+
+    u0, obs_idx, y_obs = generate_ode_dataset(n_experiments=64, n_obs=10, n_steps=30, t1=5.0)
+    m = ODEModel(lotka_volterra_rhs, u0, 0.0, 5.0, 30, obs_idx, y_obs, 0.1, device=dev)
+    rng = np.random.RandomState(83)
+    theta_c = np.abs(rng.normal([0.8, 0.3, 0.6, 0.2], 0.05, size=(8, 4))).T  # [4, 8]
+    logps, G = m.logp_grad_batched(theta_c)
+    for c in [0, 3, 7]:
+        l_ref, (g_ref,) = m(theta_c[:, c])
+        np.testing.assert_allclose(float(logps[c]), float(l_ref), rtol=1e-12)
+        np.testing.assert_allclose(
+            G[:, c].cpu().numpy(), np.asarray(g_ref), rtol=1e-10
+        )
