@@ -1,6 +1,7 @@
 // Native federated worker daemon (C++, no Python on the serving path).
 //
-// Serves a Gaussian linear-regression shard's fused logp+grad over the
+// Serves a private data shard's fused logp+grad (gaussian linear, logistic
+// GLM, or Lotka-Volterra ODE adjoint -- --model) over the
 // framework's fast transport (FEDS1 frames; payloads are the same
 // protobuf-encoded InputArrays/OutputArrays as the gRPC edge -- see
 // pytensor_federated_amd/fastsock.py and rpc.py).  The compute path is the
@@ -23,6 +24,7 @@
 #include <dlfcn.h>
 #include <netinet/in.h>
 #include <netinet/tcp.h>
+#include <math.h>
 #include <signal.h>
 #include <stdio.h>
 #include <stdlib.h>
@@ -182,26 +184,36 @@ typedef int (*eval_fn_t)(const void*, const void*, long long, double, double,
                          unsigned long long);
 typedef int (*logistic_fn_t)(const void*, const void*, long long, int,
                              const float*, double*, float*, long long, int, void*);
+typedef int (*ode_fn_t)(const double*, const double*, const int*, int, int,
+                        double, double, const double*, double*, double*, void*);
 typedef void* (*host_alloc_fn_t)(long long);
 typedef void* (*dev_alloc_fn_t)(long long);
 
 enum { FED_F32 = 0, FED_F64 = 1, FED_BF16 = 2 };
+enum { MODEL_LINEAR = 0, MODEL_LOGISTIC = 1, MODEL_ODE = 2 };
 
 struct Worker {
     void* x_dev = nullptr;
     void* y_dev = nullptr;
     long long n = 0;
-    int K = 0;                    // 0 = gaussian linear; >0 = logistic [N][K]
+    int model = MODEL_LINEAR;
+    int K = 0;                    // logistic: feature count [N][K]
     double sigma = 0.4;
     int dtype = FED_BF16;
-    double* out_dev = nullptr;    // fp64[3] or fp64[1+K]
+    double* out_dev = nullptr;    // fp64[3] / fp64[1+K] / fp64[5]
     float* beta_dev = nullptr;    // logistic: f32[K]
     double* ws_dev = nullptr;
     float* ws_f32 = nullptr;      // logistic grad slab
     double* mailbox = nullptr;    // gaussian result path
-    std::vector<double> out_host; // logistic result readback
+    std::vector<double> out_host; // logistic/ode result readback
     eval_fn_t eval = nullptr;
     logistic_fn_t eval_logistic = nullptr;
+    ode_fn_t eval_ode = nullptr;
+    // ODE (Lotka-Volterra family; see ode_lv.hip): u0=x_dev, y=y_dev
+    int ode_B = 0, ode_steps = 0;
+    double ode_h = 0.0, ode_logp_const = 0.0;
+    int* obs_dev = nullptr;       // int32[n_steps+1] step->obs row (-1 none)
+    double* theta_dev = nullptr;  // f64[4]
     unsigned long long seq = 0;
     int n_clients = 0;
 };
@@ -247,6 +259,63 @@ static bool load_logistic_shard(Worker& w, const char* path) {
         fprintf(stderr, "device upload failed\n");
         return false;
     }
+    return true;
+}
+
+// ODE shard: [i64 B][i64 n_steps][i64 n_obs][f64 h][f64 sigma]
+//            [u0: B*2 f64][obs_idx: n_obs i64][y: n_obs*B*2 f64]
+// (written by demo_node / test_native_worker; mirrors models/ode.py inputs)
+static bool load_ode_shard(Worker& w, const char* path) {
+    FILE* f = fopen(path, "rb");
+    if (!f) { fprintf(stderr, "cannot open %s\n", path); return false; }
+    long long B = 0, n_steps = 0, n_obs = 0;
+    double h = 0, sigma = 0;
+    if (fread(&B, 8, 1, f) != 1 || fread(&n_steps, 8, 1, f) != 1 ||
+        fread(&n_obs, 8, 1, f) != 1 || fread(&h, 8, 1, f) != 1 ||
+        fread(&sigma, 8, 1, f) != 1 || B <= 0 || n_steps <= 0 || n_obs <= 0) {
+        fclose(f);
+        fprintf(stderr, "bad ode shard header\n");
+        return false;
+    }
+    std::vector<double> u0(B * 2), y(n_obs * B * 2);
+    std::vector<long long> obs_idx(n_obs);
+    if (fread(u0.data(), 8, B * 2, f) != (size_t)(B * 2) ||
+        fread(obs_idx.data(), 8, n_obs, f) != (size_t)n_obs ||
+        fread(y.data(), 8, n_obs * B * 2, f) != (size_t)(n_obs * B * 2)) {
+        fclose(f);
+        fprintf(stderr, "truncated ode shard\n");
+        return false;
+    }
+    fclose(f);
+    std::vector<int> obs_of_step(n_steps + 1, -1);
+    for (long long j = 0; j < n_obs; ++j) {
+        if (obs_idx[j] < 0 || obs_idx[j] > n_steps) {
+            fprintf(stderr, "obs index %lld out of range\n", obs_idx[j]);
+            return false;
+        }
+        obs_of_step[obs_idx[j]] = (int)j;
+    }
+    w.ode_B = (int)B;
+    w.ode_steps = (int)n_steps;
+    w.ode_h = h;
+    w.sigma = sigma;
+    w.n = B;
+    const double n_vals = (double)(n_obs * B * 2);
+    w.ode_logp_const = -0.5 * n_vals * log(2.0 * M_PI * sigma * sigma);
+    if (hipMalloc(&w.x_dev, B * 2 * 8) != hipSuccess ||
+        hipMalloc(&w.y_dev, n_obs * B * 2 * 8) != hipSuccess ||
+        hipMalloc((void**)&w.obs_dev, (n_steps + 1) * 4) != hipSuccess ||
+        hipMalloc((void**)&w.theta_dev, 4 * 8) != hipSuccess ||
+        hipMalloc((void**)&w.out_dev, 5 * 8) != hipSuccess ||
+        hipMalloc((void**)&w.ws_dev, (n_steps + 1) * B * 2 * 8) != hipSuccess ||
+        hipMemcpy(w.x_dev, u0.data(), B * 2 * 8, hipMemcpyHostToDevice) != hipSuccess ||
+        hipMemcpy(w.y_dev, y.data(), n_obs * B * 2 * 8, hipMemcpyHostToDevice) != hipSuccess ||
+        hipMemcpy(w.obs_dev, obs_of_step.data(), (n_steps + 1) * 4,
+                  hipMemcpyHostToDevice) != hipSuccess) {
+        fprintf(stderr, "device upload failed\n");
+        return false;
+    }
+    w.out_host.resize(5);
     return true;
 }
 
@@ -343,7 +412,7 @@ static void serve_client(Worker& w, int fd) {
         payload.resize(ln);
         if (ln && !read_exact(fd, payload.data(), ln)) break;
         static std::mutex eval_mu;
-        if (hdr[0] == 0x01 && w.K == 0) {  // Evaluate: gaussian linear
+        if (hdr[0] == 0x01 && w.model == MODEL_LINEAR) {  // Evaluate: gaussian linear
             std::vector<ParsedArray> items;
             std::string uuid;
             if (!parse_input_arrays(payload.data(), ln, items, uuid) || items.size() != 2) {
@@ -371,6 +440,37 @@ static void serve_client(Worker& w, int fd) {
             put_len_field(out, 1, encode_f64_scalar(res[0]));  // logp
             put_len_field(out, 1, encode_f64_scalar(res[1]));  // d/da
             put_len_field(out, 1, encode_f64_scalar(res[2]));  // d/db
+            put_len_field(out, 2, uuid);
+            if (!write_frame(fd, 0x81, out)) break;
+        } else if (hdr[0] == 0x01 && w.model == MODEL_ODE) {  // theta[4] f64 in
+            std::vector<ParsedArray> items;
+            std::string uuid;
+            if (!parse_input_arrays(payload.data(), ln, items, uuid) || items.size() != 1 ||
+                items[0].dtype != "float64" || items[0].data.size() != 4 * 8) {
+                write_frame(fd, 0xFF, "expected one float64 theta[4] input");
+                continue;
+            }
+            std::string out;
+            {
+                std::lock_guard<std::mutex> lock(eval_mu);
+                int rc = (int)hipMemcpy(w.theta_dev, items[0].data.data(), 4 * 8,
+                                        hipMemcpyHostToDevice);
+                if (rc == 0)
+                    rc = w.eval_ode((const double*)w.x_dev, (const double*)w.y_dev,
+                                    w.obs_dev, w.ode_steps, w.ode_B, w.ode_h,
+                                    w.sigma, w.theta_dev, w.ws_dev, w.out_dev,
+                                    nullptr);
+                if (rc == 0) rc = (int)hipDeviceSynchronize();
+                if (rc == 0)
+                    rc = (int)hipMemcpy(w.out_host.data(), w.out_dev, 5 * 8,
+                                        hipMemcpyDeviceToHost);
+                if (rc != 0) {
+                    write_frame(fd, 0xFF, "ode eval failed");
+                    continue;
+                }
+                put_len_field(out, 1, encode_f64_scalar(w.out_host[0] + w.ode_logp_const));
+                put_len_field(out, 1, encode_f64_vector(&w.out_host[1], 4));
+            }
             put_len_field(out, 2, uuid);
             if (!write_frame(fd, 0x81, out)) break;
         } else if (hdr[0] == 0x01) {  // Evaluate: logistic GLM, beta[K] f64 in
@@ -428,20 +528,25 @@ static void serve_client(Worker& w, int fd) {
 int main(int argc, char** argv) {
     int port = 9600;
     const char* data_path = nullptr;
-    bool logistic = false;
     Worker w;
     for (int i = 1; i < argc - 1; ++i) {
         if (!strcmp(argv[i], "--port")) port = atoi(argv[++i]);
         else if (!strcmp(argv[i], "--data")) data_path = argv[++i];
         else if (!strcmp(argv[i], "--sigma")) w.sigma = atof(argv[++i]);
-        else if (!strcmp(argv[i], "--model")) logistic = !strcmp(argv[++i], "logistic");
+        else if (!strcmp(argv[i], "--model")) {
+            const char* m = argv[++i];
+            w.model = !strcmp(m, "logistic") ? MODEL_LOGISTIC
+                    : !strcmp(m, "ode")      ? MODEL_ODE
+                                             : MODEL_LINEAR;
+        }
         else if (!strcmp(argv[i], "--dtype")) {
             const char* d = argv[++i];
             w.dtype = !strcmp(d, "f64") ? FED_F64 : !strcmp(d, "f32") ? FED_F32 : FED_BF16;
         }
     }
     if (!data_path) {
-        fprintf(stderr, "usage: fed_worker --port P --data shard.bin [--sigma S] [--dtype bf16|f32|f64]\n");
+        fprintf(stderr, "usage: fed_worker --port P --data shard.bin "
+                        "[--model linear|logistic|ode] [--sigma S] [--dtype bf16|f32|f64]\n");
         return 2;
     }
     signal(SIGPIPE, SIG_IGN);
@@ -456,12 +561,13 @@ int main(int argc, char** argv) {
     }
     w.eval = (eval_fn_t)dlsym(lib, "fed_gaussian_linear_eval");
     w.eval_logistic = (logistic_fn_t)dlsym(lib, "fed_logistic_glm");
+    w.eval_ode = (ode_fn_t)dlsym(lib, "fed_ode_lv_eval");
     host_alloc_fn_t host_alloc = (host_alloc_fn_t)dlsym(lib, "fed_host_alloc");
-    if (!w.eval || !w.eval_logistic || !host_alloc) {
+    if (!w.eval || !w.eval_logistic || !w.eval_ode || !host_alloc) {
         fprintf(stderr, "missing symbols in %s\n", lib_path.c_str());
         return 2;
     }
-    if (logistic) {
+    if (w.model == MODEL_LOGISTIC) {
         if (!load_logistic_shard(w, data_path)) return 2;
         w.out_host.resize(1 + w.K);
         if (hipMalloc((void**)&w.out_dev, (1 + w.K) * 8) != hipSuccess ||
@@ -470,12 +576,15 @@ int main(int argc, char** argv) {
             fprintf(stderr, "device alloc failed\n");
             return 2;
         }
+    } else if (w.model == MODEL_ODE) {
+        if (!load_ode_shard(w, data_path)) return 2;
     } else if (!load_shard(w, data_path)) {
         return 2;
     }
-    if (!logistic && (hipMalloc(&w.out_dev, 3 * 8) != hipSuccess ||
-        hipMalloc(&w.ws_dev, (72 + 3 * 2048) * 8) != hipSuccess ||
-        hipMemset(w.ws_dev, 0, (72 + 3 * 2048) * 8) != hipSuccess)) {
+    if (w.model == MODEL_LINEAR &&
+        (hipMalloc(&w.out_dev, 3 * 8) != hipSuccess ||
+         hipMalloc(&w.ws_dev, (72 + 3 * 2048) * 8) != hipSuccess ||
+         hipMemset(w.ws_dev, 0, (72 + 3 * 2048) * 8) != hipSuccess)) {
         fprintf(stderr, "device alloc failed\n");
         return 2;
     }

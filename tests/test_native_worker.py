@@ -80,6 +80,56 @@ def test_cpp_worker_serves_kernel_evals(tmp_path):
 
 
 @pytest.mark.timeout(300)
+def test_cpp_worker_serves_ode(tmp_path):
+    from pytensor_federated_amd.common import LogpGradServiceClient
+    from pytensor_federated_amd.models import (
+        ODEModel,
+        generate_ode_dataset,
+        lotka_volterra_rhs,
+    )
+
+    import torch
+
+    u0, obs_idx, y = generate_ode_dataset(
+        n_experiments=64, n_obs=20, n_steps=100, sigma=0.1, seed=53
+    )
+    shard = tmp_path / "ode.bin"
+    with open(shard, "wb") as f:
+        f.write(struct.pack("<qqqdd", u0.shape[0], 100, len(obs_idx), 10.0 / 100, 0.1))
+        f.write(np.asarray(u0, dtype=np.float64).tobytes())
+        f.write(np.asarray(obs_idx, dtype=np.int64).tobytes())
+        f.write(np.asarray(y, dtype=np.float64).tobytes())
+
+    env = dict(os.environ, FEDOPS_LIB=str(LIB))
+    proc = subprocess.Popen(
+        [str(WORKER), "--port", str(PORT + 4), "--data", str(shard), "--model", "ode"],
+        env=env, stderr=subprocess.PIPE,
+    )
+    try:
+        _wait_tcp(PORT + 4)
+        client = LogpGradServiceClient("127.0.0.1", PORT + 4, transport="fast")
+        theta = np.array([0.8, 0.3, 0.6, 0.2])
+        logp, (grad,) = client.evaluate(theta)
+        ref_model = ODEModel(
+            lotka_volterra_rhs, u0, 0.0, 10.0, 100, obs_idx, y, sigma=0.1,
+            device="cuda:0", use_kernels=True,
+        )
+        logp_ref, (g_ref,) = ref_model.logp_grad(torch.as_tensor(theta))
+        np.testing.assert_allclose(float(logp), float(logp_ref), rtol=1e-10)
+        np.testing.assert_allclose(grad, g_ref.cpu().numpy(), rtol=1e-9)
+        # second theta exercises the theta re-upload path
+        theta2 = np.array([0.9, 0.25, 0.55, 0.22])
+        logp2, (grad2,) = client.evaluate(theta2)
+        logp2_ref, (g2_ref,) = ref_model.logp_grad(torch.as_tensor(theta2))
+        np.testing.assert_allclose(float(logp2), float(logp2_ref), rtol=1e-10)
+        np.testing.assert_allclose(grad2, g2_ref.cpu().numpy(), rtol=1e-9)
+        del client
+    finally:
+        proc.terminate()
+        proc.wait(timeout=10)
+
+
+@pytest.mark.timeout(300)
 def test_cpp_worker_serves_logistic(tmp_path):
     from pytensor_federated_amd.common import LogpGradServiceClient
     from pytensor_federated_amd.models import LogisticGLMModel, generate_logistic_dataset
