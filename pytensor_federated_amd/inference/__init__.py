@@ -13,3 +13,4 @@ from .mala import sample_mala_batched  # noqa: F401
 from .map import find_map  # noqa: F401
 from .mcmc import Metropolis, sample_metropolis  # noqa: F401
 from .nuts import NUTS, sample_nuts  # noqa: F401
+from .nuts_batched import sample_nuts_batched  # noqa: F401

@@ -128,7 +128,7 @@ class NUTS:
             s_prime = log_u < joint + _DELTA_MAX
             if not s_prime:
                 self.n_divergent += 1
-            alpha = min(1.0, np.exp(joint - joint0))
+            alpha = min(1.0, np.exp(min(joint - joint0, 0.0)))
             return (q1, p1, grad1, q1, p1, grad1, q1, logp1, grad1,
                     n_prime, s_prime, alpha, 1)
         (q_minus, p_minus, grad_minus, q_plus, p_plus, grad_plus, q_prop,
