@@ -25,6 +25,7 @@ def main():
     parser.add_argument("--experiments", type=int, default=256)
     parser.add_argument("--draws", type=int, default=200)
     parser.add_argument("--tune", type=int, default=400)
+    parser.add_argument("--mass", default="diag", choices=["diag", "dense"])
     args = parser.parse_args()
 
     import torch
@@ -58,7 +59,7 @@ def main():
     t0 = time.perf_counter()
     chain, stats = sample_nuts_batched(
         batched, init, draws=args.draws, tune=args.tune, step_size=5e-4,
-        seed=87, max_depth=8,
+        seed=87, max_depth=8, mass=args.mass,
     )
     wall = time.perf_counter() - t0
     post_mean = chain.mean(axis=(0, 2))
@@ -70,13 +71,21 @@ def main():
         "rounds": stats["rounds"],
         "leapfrogs": stats["leapfrogs"],
         "amortization": stats["leapfrogs"] / max(stats["rounds"], 1),
+        "accept_stat_range": [float(min(stats["accept_stat"])),
+                              float(max(stats["accept_stat"]))],
+        "chain_rel_sd": [float(v) for v in
+                         (chain.std(axis=0).mean(axis=1) / np.abs(post_mean))],
+        "between_rel_sd": [float(v) for v in
+                           (chain.mean(axis=0).std(axis=1) / np.abs(post_mean))],
+        "step_size_range": [float(min(stats["step_sizes"])),
+                            float(max(stats["step_sizes"]))],
         "posterior_mean": [float(v) for v in post_mean],
         "max_rel_err_vs_truth": float(np.max(np.abs(post_mean / theta_true - 1))),
         "split_rhat_max": float(max(
             split_rhat(chain[:, k, :].T) for k in range(4)
         )),
         "config": {"chains": C, "experiments": args.experiments,
-                   "model": "lotka_volterra_ode_adjoint", "dtype": "f64",
+                   "model": "lotka_volterra_ode_adjoint", "dtype": "f64", "mass": args.mass,
                    "kernel": "k_lv_forward_batched/k_lv_adjoint_batched"},
     }))
 

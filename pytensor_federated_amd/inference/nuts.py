@@ -124,6 +124,8 @@ class NUTS:
         if j == 0:
             q1, p1, logp1, grad1 = self._leapfrog(q, p, grad, v * self.step_size)
             joint = logp1 - self._kinetic(p1)
+            if not np.isfinite(joint):
+                joint = -np.inf  # out of support / overflow: reject cleanly
             n_prime = 1 if log_u <= joint else 0
             s_prime = log_u < joint + _DELTA_MAX
             if not s_prime:
@@ -163,12 +165,29 @@ class NUTS:
         eta = 1.0 / (m + self._t0)
         self._h_bar = (1 - eta) * self._h_bar + eta * (self.target_accept - a_stat)
         log_eps = self._mu - np.sqrt(m) / self._gamma * self._h_bar
+        # bound the early-iteration overshoot (dual averaging can run away
+        # double-exponentially while the acceptance statistic saturates)
+        log_eps = float(np.clip(log_eps, self._mu - 20.0, self._mu + 20.0))
         w = m ** (-self._kappa)
         self._log_eps_bar = w * log_eps + (1 - w) * self._log_eps_bar
         self.step_size = float(np.exp(log_eps))
 
     def freeze_step_size(self) -> None:
         self.step_size = float(np.exp(self._log_eps_bar))
+
+    def reset_step_size_adaptation(self, step_size: float) -> None:
+        """Restart dual averaging anchored at a fresh step size.
+
+        MUST be called when the mass matrix changes: eps is measured in
+        metric-whitened units, so a metric update invalidates both the
+        current eps and the averaging anchor ``mu`` (Stan restarts its
+        step-size adaptation at every metric-window boundary for the same
+        reason)."""
+        self.step_size = float(step_size)
+        self._mu = np.log(10 * self.step_size)
+        self._log_eps_bar = 0.0
+        self._h_bar = 0.0
+        self._adapt_count = 0
 
 
 def sample_nuts(
@@ -200,5 +219,8 @@ def sample_nuts(
                 var = np.var(np.stack(window[len(window) // 2 :]), axis=0)
                 sampler.inv_mass = np.maximum(var, 1e-10)
                 window.clear()
+                # eps units changed with the metric: restart dual averaging
+                # at a conservative whitened-units step
+                sampler.reset_step_size_adaptation(0.25)
     sampler.freeze_step_size()
     return [sampler.step() for _ in range(draws)]

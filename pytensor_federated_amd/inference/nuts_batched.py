@@ -55,20 +55,49 @@ class _ChainNUTS:
         self.logp: float = 0.0
         self.grad: np.ndarray = np.zeros_like(self.q)
         self._last_accept_stat = 1.0
+        # dense mass (Stan-style metric): M^-1 = Sigma (posterior covariance
+        # estimate); None = diagonal metric via inv_mass
+        self.mass_sigma: Optional[np.ndarray] = None
+        self._mass_chol: Optional[np.ndarray] = None
+
+    def _vel(self, p):
+        """dq/dt = M^-1 p."""
+        if self.mass_sigma is None:
+            return self.inv_mass * p
+        return self.mass_sigma @ p
 
     def _kinetic(self, p):
-        return 0.5 * float(np.sum(self.inv_mass * p * p))
+        if self.mass_sigma is None:
+            # keep the exact summation order of nuts.NUTS._kinetic: the
+            # C == 1 bit-identity test depends on it
+            return 0.5 * float(np.sum(self.inv_mass * p * p))
+        return 0.5 * float(p @ self._vel(p))
+
+    def _sample_momentum(self):
+        z = self.rng.normal(size=self.q.shape)
+        if self.mass_sigma is None:
+            return z / np.sqrt(self.inv_mass)
+        # p ~ N(0, Sigma^-1): with Sigma = L L^T, p = L^-T z
+        return np.linalg.solve(self._mass_chol.T, z)
+
+    def set_dense_mass(self, sigma: np.ndarray) -> None:
+        self.mass_sigma = sigma
+        self._mass_chol = np.linalg.cholesky(sigma)
 
     def _leapfrog(self, q, p, grad, eps):
         p = p + 0.5 * eps * grad
-        q = q + eps * self.inv_mass * p
+        if self.mass_sigma is None:
+            # exact expression grouping of nuts.NUTS._leapfrog (bit-identity)
+            q = q + eps * self.inv_mass * p
+        else:
+            q = q + eps * (self.mass_sigma @ p)
         logp, grad = yield q
         p = p + 0.5 * eps * grad
         return q, p, logp, grad
 
     def transition(self):
         q0, logp0, grad0 = self.q, self.logp, self.grad
-        p0 = self.rng.normal(size=q0.shape) / np.sqrt(self.inv_mass)
+        p0 = self._sample_momentum()
         joint0 = logp0 - self._kinetic(p0)
         log_u = joint0 + np.log(self.rng.uniform())
 
@@ -95,8 +124,8 @@ class _ChainNUTS:
             alpha_sum += a
             n_alpha += na
             dq = q_plus - q_minus
-            s = s_prime and (dq @ (self.inv_mass * p_minus) >= 0) and (
-                dq @ (self.inv_mass * p_plus) >= 0
+            s = s_prime and (dq @ self._vel(p_minus) >= 0) and (
+                dq @ self._vel(p_plus) >= 0
             )
             j += 1
 
@@ -108,6 +137,8 @@ class _ChainNUTS:
             q1, p1, logp1, grad1 = yield from self._leapfrog(
                 q, p, grad, v * self.step_size)
             joint = logp1 - self._kinetic(p1)
+            if not np.isfinite(joint):
+                joint = -np.inf  # out of support / overflow: reject cleanly
             n_prime = 1 if log_u <= joint else 0
             s_prime = log_u < joint + _DELTA_MAX
             if not s_prime:
@@ -133,8 +164,8 @@ class _ChainNUTS:
             a += a2
             na += na2
             dq = q_plus - q_minus
-            s_prime = s_pp and (dq @ (self.inv_mass * p_minus) >= 0) and (
-                dq @ (self.inv_mass * p_plus) >= 0
+            s_prime = s_pp and (dq @ self._vel(p_minus) >= 0) and (
+                dq @ self._vel(p_plus) >= 0
             )
         return (q_minus, p_minus, grad_minus, q_plus, p_plus, grad_plus,
                 q_prop, logp_prop, grad_prop, n_prime, s_prime, a, na)
@@ -147,12 +178,22 @@ class _ChainNUTS:
         self._h_bar = (1 - eta) * self._h_bar + eta * (
             self.target_accept - self._last_accept_stat)
         log_eps = self._mu - np.sqrt(m) / self._gamma * self._h_bar
+        log_eps = float(np.clip(log_eps, self._mu - 20.0, self._mu + 20.0))
         w = m ** (-self._kappa)
         self._log_eps_bar = w * log_eps + (1 - w) * self._log_eps_bar
         self.step_size = float(np.exp(log_eps))
 
     def freeze_step_size(self):
         self.step_size = float(np.exp(self._log_eps_bar))
+
+    def reset_step_size_adaptation(self, step_size: float) -> None:
+        """Restart dual averaging at a fresh (whitened-units) step size --
+        required after every metric update; see nuts.NUTS for rationale."""
+        self.step_size = float(step_size)
+        self._mu = np.log(10 * self.step_size)
+        self._log_eps_bar = 0.0
+        self._h_bar = 0.0
+        self._adapt_count = 0
 
 
 def sample_nuts_batched(
@@ -165,6 +206,7 @@ def sample_nuts_batched(
     target_accept: float = 0.8,
     seed: Optional[int] = None,
     adapt_mass: bool = True,
+    mass: str = "diag",
     max_depth: int = _MAX_DEPTH,
 ) -> Tuple[np.ndarray, dict]:
     """Run C lockstep NUTS chains over one batched evaluator.
@@ -177,6 +219,10 @@ def sample_nuts_batched(
     init : array [K, C]
         Initial states of the C chains (chain c uses seed ``seed + c``, so
         C == 1 reproduces ``sample_nuts(..., seed=seed)`` exactly).
+    mass : "diag" (default, matches ``sample_nuts``) or "dense"
+        "dense" adapts a full covariance metric from the tuning window --
+        required for strongly correlated posteriors (e.g. ODE parameters),
+        where a diagonal metric mixes arbitrarily slowly.
 
     Returns
     -------
@@ -209,6 +255,7 @@ def sample_nuts_batched(
 
     total = tune + draws
     samples = np.empty((draws, K, C))
+    accept_stats = [[] for _ in range(C)]
     n_done = [0] * C
     windows: List[List[np.ndarray]] = [[] for _ in range(C)]
     mass_update_at = int(tune * 0.6)
@@ -228,14 +275,26 @@ def sample_nuts_batched(
             if adapt_mass:
                 windows[c].append(ch.q.copy())
                 if i == mass_update_at and len(windows[c]) > 10:
-                    w = windows[c]
-                    var = np.var(np.stack(w[len(w) // 2:]), axis=0)
-                    ch.inv_mass = np.maximum(var, 1e-10)
+                    w = np.stack(windows[c][len(windows[c]) // 2:])
+                    if mass == "dense":
+                        # shrink toward the diagonal (Stan-style regularized
+                        # covariance metric) so the Cholesky stays stable
+                        n_w = w.shape[0]
+                        cov = np.atleast_2d(np.cov(w.T))
+                        lam = n_w / (n_w + 5.0)
+                        diag = np.diag(np.maximum(np.diag(cov), 1e-12))
+                        ch.set_dense_mass(lam * cov + (1 - lam) * diag
+                                          + 1e-12 * np.eye(cov.shape[0]))
+                    else:
+                        var = np.var(w, axis=0)
+                        ch.inv_mass = np.maximum(var, 1e-10)
                     windows[c].clear()
+                    ch.reset_step_size_adaptation(0.25)
             if n_done[c] == tune:
                 ch.freeze_step_size()
         else:
             samples[i - tune, :, c] = ch.q
+            accept_stats[c].append(ch._last_accept_stat)
         if n_done[c] >= total:
             gens[c] = None
             return None
@@ -265,4 +324,6 @@ def sample_nuts_batched(
         "leapfrogs": leapfrogs,
         "step_sizes": [ch.step_size for ch in chains],
         "divergences": [ch.n_divergent for ch in chains],
+        "accept_stat": [float(np.mean(a)) if a else float("nan")
+                        for a in accept_stats],
     }

@@ -151,6 +151,42 @@ class TestNUTSBatched:
         assert stats["leapfrogs"] > 2.0 * stats["rounds"]
         assert all(d == 0 for d in stats["divergences"])
 
+    def test_dense_mass_on_correlated_posterior(self):
+        """Dense (full-covariance) metric on a sharp rho=0.99 posterior:
+        converged R-hat and materially higher ESS than the diagonal metric
+        buys on the same budget."""
+        from pytensor_federated_amd.inference import sample_nuts_batched
+        from pytensor_federated_amd.inference.diagnostics import (
+            effective_sample_size,
+            split_rhat,
+        )
+
+        mu = np.array([0.8, 0.3, 0.6, 0.2])
+        sd = np.array([1e-3, 3e-4, 1e-3, 3e-4])
+        R = np.eye(4)
+        R[0, 1] = R[1, 0] = 0.99
+        R[2, 3] = R[3, 2] = 0.97
+        prec = np.linalg.inv(np.outer(sd, sd) * R)
+
+        def batched(theta):
+            d = theta - mu[:, None]
+            logp = -0.5 * np.einsum("kb,kj,jb->b", d, prec, d)
+            return logp, -prec @ d
+
+        C = 8
+        init = np.tile(mu[:, None], (1, C)) * (
+            1 + 0.005 * np.random.RandomState(86).standard_normal((4, C))
+        )
+        chain, stats = sample_nuts_batched(
+            batched, init, draws=150, tune=300, step_size=5e-4, seed=87,
+            max_depth=8, mass="dense",
+        )
+        assert max(split_rhat(chain[:, k, :].T) for k in range(4)) < 1.05
+        assert min(
+            effective_sample_size(chain[:, k, :].T.reshape(-1)) for k in range(4)
+        ) > 300
+        np.testing.assert_allclose(chain.mean(axis=(0, 2)), mu, atol=5e-4)
+
     def test_batched_linear_model_matches_conjugate(self):
         """Lockstep chains over a model's python-level batched evaluator
         reproduce the closed-form flat-prior posterior."""
