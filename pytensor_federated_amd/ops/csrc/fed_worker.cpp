@@ -101,6 +101,11 @@ static std::string encode_f64_vector(const double* values, int n) {
     return msg;
 }
 
+static void put_float_field(std::string& out, int field, float v) {
+    put_varint(out, (field << 3) | 5);  // wire type 5: fixed32
+    out.append((const char*)&v, 4);
+}
+
 struct ParsedArray {
     std::vector<unsigned char> data;
     std::string dtype;
@@ -512,9 +517,39 @@ static void serve_client(Worker& w, int fd) {
             put_len_field(out, 2, uuid);
             if (!write_frame(fd, 0x81, out)) break;
         } else if (hdr[0] == 0x02) {  // GetLoad
+            // GPU-first telemetry, mirroring the Python edge's
+            // service.determine_load (service.py:141-155): percent_cpu
+            // carries GPU busy %, percent_ram carries VRAM use %.
             std::string out;
             put_varint(out, (1 << 3) | 0);  // n_clients
             put_varint(out, (unsigned long long)w.n_clients);
+            float busy = -1.0f;
+            for (int card = 0; card < 8 && busy < 0; ++card) {
+                char path[64];
+                snprintf(path, sizeof(path),
+                         "/sys/class/drm/card%d/device/gpu_busy_percent", card);
+                FILE* f = fopen(path, "r");
+                if (f) {
+                    int v;
+                    if (fscanf(f, "%d", &v) == 1) busy = (float)v;
+                    fclose(f);
+                }
+            }
+            if (busy < 0) {  // no amdgpu sysfs: host loadavg fallback
+                double la = 0;
+                if (getloadavg(&la, 1) == 1) {
+                    long ncpu = sysconf(_SC_NPROCESSORS_ONLN);
+                    busy = (float)(100.0 * la / (ncpu > 0 ? ncpu : 1));
+                } else {
+                    busy = 0.0f;
+                }
+            }
+            float vram = 0.0f;
+            size_t free_b = 0, total_b = 0;
+            if (hipMemGetInfo(&free_b, &total_b) == hipSuccess && total_b)
+                vram = (float)(100.0 * (1.0 - (double)free_b / (double)total_b));
+            put_float_field(out, 2, busy);
+            put_float_field(out, 3, vram);
             if (!write_frame(fd, 0x82, out)) break;
         } else {
             write_frame(fd, 0xFF, "unknown frame type");

@@ -50,6 +50,18 @@ def test_cpp_worker_serves_kernel_evals(tmp_path):
     )
     try:
         _wait_tcp(PORT)
+        # GetLoad telemetry from the C++ worker (before the client opens its
+        # persistent connection: asyncio.run uses its own event loop)
+        import asyncio
+
+        from pytensor_federated_amd.service import get_load_async
+
+        load = asyncio.run(get_load_async("127.0.0.1", PORT, transport="fast"))
+        assert load is not None
+        assert 0.0 <= load.percent_cpu <= 100.0
+        # hipMemGetInfo granularity can round a small shard to 0% on 288 GB
+        assert 0.0 <= load.percent_ram <= 100.0
+
         client = LogpGradServiceClient("127.0.0.1", PORT, transport="fast")
         logp, (ga, gb) = client.evaluate(1.5, 0.5)
 
