@@ -217,6 +217,45 @@ class TestNUTSBatched:
             assert abs(flat.std(axis=0)[k] - post_sd[k]) < 0.4 * post_sd[k]
 
 
+class TestConstrainedSupport:
+    def test_nuts_rejects_out_of_support_cleanly(self):
+        """Hard support boundary (logp = -inf for x <= 0): leapfrog steps
+        that cross it must become clean rejections (non-finite Hamiltonian
+        guard), never crashes or out-of-support draws."""
+        def logp_grad(x):
+            v = float(x[0])
+            if v <= 0:
+                return np.array(-np.inf), [np.zeros(1)]
+            # Gamma(3, 1): logp = 2 log x - x
+            return np.array(2 * np.log(v) - v), [np.array([2.0 / v - 1.0])]
+
+        chain = sample_nuts(
+            logp_grad, [np.array([0.1])], draws=800, tune=400, seed=21
+        )
+        samples = np.array([float(d[0][0]) for d in chain])
+        assert np.all(samples > 0)
+        # Gamma(3,1): mean 3, sd sqrt(3)
+        assert abs(samples.mean() - 3.0) < 0.5
+        assert abs(samples.std() - np.sqrt(3.0)) < 0.6
+
+    def test_nuts_batched_rejects_out_of_support_cleanly(self):
+        from pytensor_federated_amd.inference import sample_nuts_batched
+
+        def batched(theta):
+            x = theta[0]
+            ok = x > 0
+            logp = np.where(ok, 2 * np.log(np.where(ok, x, 1.0)) - x, -np.inf)
+            grad = np.where(ok, 2.0 / np.where(ok, x, 1.0) - 1.0, 0.0)
+            return logp, grad[None, :]
+
+        chain, stats = sample_nuts_batched(
+            batched, np.full((1, 4), 0.1), draws=500, tune=300, seed=22
+        )
+        assert np.all(chain > 0)
+        pooled = chain.reshape(-1)
+        assert abs(pooled.mean() - 3.0) < 0.5
+
+
 class TestMALABatched:
     def test_recovers_gaussian_all_chains(self):
         from pytensor_federated_amd.inference import sample_mala_batched
