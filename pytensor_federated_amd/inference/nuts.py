@@ -64,7 +64,26 @@ class NUTS:
         self._kappa = 0.75
         self._adapt_count = 0
         self.n_divergent = 0
+        # dense metric (M^-1 = Sigma, a posterior-covariance estimate);
+        # None = diagonal metric via inv_mass
+        self.mass_sigma = None
+        self._mass_chol = None
         self._logp, self._grad = self._eval(self.q)
+
+    def set_dense_mass(self, sigma: np.ndarray) -> None:
+        self.mass_sigma = sigma
+        self._mass_chol = np.linalg.cholesky(sigma)
+
+    def _vel(self, p: np.ndarray) -> np.ndarray:
+        if self.mass_sigma is None:
+            return self.inv_mass * p
+        return self.mass_sigma @ p
+
+    def _sample_momentum(self) -> np.ndarray:
+        z = self.rng.normal(size=self.q.shape)
+        if self.mass_sigma is None:
+            return z / np.sqrt(self.inv_mass)
+        return np.linalg.solve(self._mass_chol.T, z)  # p ~ N(0, Sigma^-1)
 
     # -- model evaluation ------------------------------------------------
     def _eval(self, q: np.ndarray) -> Tuple[float, np.ndarray]:
@@ -73,18 +92,23 @@ class NUTS:
 
     def _leapfrog(self, q, p, grad, eps):
         p = p + 0.5 * eps * grad
-        q = q + eps * self.inv_mass * p
+        if self.mass_sigma is None:
+            q = q + eps * self.inv_mass * p
+        else:
+            q = q + eps * (self.mass_sigma @ p)
         logp, grad = self._eval(q)
         p = p + 0.5 * eps * grad
         return q, p, logp, grad
 
     def _kinetic(self, p: np.ndarray) -> float:
-        return 0.5 * float(np.sum(self.inv_mass * p * p))
+        if self.mass_sigma is None:
+            return 0.5 * float(np.sum(self.inv_mass * p * p))
+        return 0.5 * float(p @ self._vel(p))
 
     # -- one NUTS transition ----------------------------------------------
     def step(self) -> List[np.ndarray]:
         q0, logp0, grad0 = self.q, self._logp, self._grad
-        p0 = self.rng.normal(size=q0.shape) / np.sqrt(self.inv_mass)
+        p0 = self._sample_momentum()
         joint0 = logp0 - self._kinetic(p0)
         log_u = joint0 + np.log(self.rng.uniform())
 
@@ -111,8 +135,8 @@ class NUTS:
             alpha_sum += a
             n_alpha += na
             dq = q_plus - q_minus
-            s = s_prime and (dq @ (self.inv_mass * p_minus) >= 0) and (
-                dq @ (self.inv_mass * p_plus) >= 0
+            s = s_prime and (dq @ self._vel(p_minus) >= 0) and (
+                dq @ self._vel(p_plus) >= 0
             )
             j += 1
 
@@ -151,8 +175,8 @@ class NUTS:
             a += a2
             na += na2
             dq = q_plus - q_minus
-            s_prime = s_pp and (dq @ (self.inv_mass * p_minus) >= 0) and (
-                dq @ (self.inv_mass * p_plus) >= 0
+            s_prime = s_pp and (dq @ self._vel(p_minus) >= 0) and (
+                dq @ self._vel(p_plus) >= 0
             )
         return (q_minus, p_minus, grad_minus, q_plus, p_plus, grad_plus,
                 q_prop, logp_prop, grad_prop, n_prime, s_prime, a, na)
@@ -200,8 +224,14 @@ def sample_nuts(
     target_accept: float = 0.8,
     seed: Optional[int] = None,
     adapt_mass: bool = True,
+    mass: str = "diag",
 ) -> List[List[np.ndarray]]:
-    """NUTS with dual-averaging step size + diagonal mass adaptation.
+    """NUTS with dual-averaging step size + mass-matrix adaptation.
+
+    ``mass="diag"`` (default) adapts a diagonal metric; ``mass="dense"``
+    adapts a full covariance metric from the tuning window (required for
+    strongly correlated posteriors).  Step-size adaptation restarts after
+    the metric update (eps lives in metric-whitened units).
 
     Returns ``draws`` samples, each a list of parameter arrays.
     """
@@ -216,8 +246,17 @@ def sample_nuts(
             window.append(sampler.q.copy())
             # one mass update mid-tune, then keep adapting step size
             if i == int(tune * 0.6) and len(window) > 10:
-                var = np.var(np.stack(window[len(window) // 2 :]), axis=0)
-                sampler.inv_mass = np.maximum(var, 1e-10)
+                w = np.stack(window[len(window) // 2 :])
+                if mass == "dense":
+                    n_w = w.shape[0]
+                    cov = np.atleast_2d(np.cov(w.T))
+                    lam = n_w / (n_w + 5.0)
+                    diag = np.diag(np.maximum(np.diag(cov), 1e-12))
+                    sampler.set_dense_mass(lam * cov + (1 - lam) * diag
+                                           + 1e-12 * np.eye(cov.shape[0]))
+                else:
+                    var = np.var(w, axis=0)
+                    sampler.inv_mass = np.maximum(var, 1e-10)
                 window.clear()
                 # eps units changed with the metric: restart dual averaging
                 # at a conservative whitened-units step

@@ -89,6 +89,21 @@ class TestNUTS:
             assert abs(samples.mean(axis=0)[k] - beta_hat[k]) < 6 * post_sd[k]
             assert abs(samples.std(axis=0)[k] - post_sd[k]) < 0.4 * post_sd[k]
 
+    def test_dense_mass_sequential(self):
+        """mass="dense" on a sharp rho=0.99 posterior (sequential sampler)."""
+        mu = np.array([0.8, 0.3])
+        sd = np.array([1e-3, 3e-4])
+        R = np.array([[1.0, 0.99], [0.99, 1.0]])
+        prec = np.linalg.inv(np.outer(sd, sd) * R)
+        chain = sample_nuts(
+            gaussian_2d_logp_grad(mu, prec),
+            [mu * 1.005],
+            draws=400, tune=300, step_size=5e-4, seed=23, mass="dense",
+        )
+        samples = np.stack([d[0] for d in chain])
+        np.testing.assert_allclose(samples.mean(axis=0), mu, atol=5e-4)
+        np.testing.assert_allclose(samples.std(axis=0), sd, rtol=0.3)
+
     def test_divergences_rare_on_gaussian(self):
         from pytensor_federated_amd.inference.nuts import NUTS
 
