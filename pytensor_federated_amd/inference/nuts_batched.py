@@ -196,6 +196,42 @@ class _ChainNUTS:
         self._adapt_count = 0
 
 
+def metric_window_ends(tune: int, base_window: int = 25,
+                       init_buffer: Optional[int] = None,
+                       term_buffer: Optional[int] = None) -> List[int]:
+    """Stan-style expanding adaptation windows: transition indices (1-based)
+    at which the metric is re-estimated and step-size adaptation restarts.
+
+    Layout: [init_buffer: step size only][w][2w][4w]...[term_buffer: step
+    size only]; the last window is extended to absorb any remainder."""
+    if init_buffer is None:
+        init_buffer = max(20, int(0.15 * tune))
+    if term_buffer is None:
+        term_buffer = max(25, int(0.1 * tune))
+    ends: List[int] = []
+    pos, w = init_buffer, base_window
+    while pos + w + term_buffer <= tune:
+        nxt = pos + w
+        if nxt + 2 * w + term_buffer > tune:
+            nxt = tune - term_buffer  # absorb the remainder
+        ends.append(nxt)
+        pos, w = nxt, w * 2
+    return ends
+
+
+def _estimate_metric(ch, w: np.ndarray, mass: str) -> None:
+    """Set ch's metric from window draws w[n, K] (shrunk toward diagonal)."""
+    if mass == "dense":
+        n_w = w.shape[0]
+        cov = np.atleast_2d(np.cov(w.T))
+        lam = n_w / (n_w + 5.0)
+        diag = np.diag(np.maximum(np.diag(cov), 1e-12))
+        ch.set_dense_mass(lam * cov + (1 - lam) * diag
+                          + 1e-12 * np.eye(cov.shape[0]))
+    else:
+        ch.inv_mass = np.maximum(np.var(w, axis=0), 1e-10)
+
+
 def sample_nuts_batched(
     batched_logp_grad: Callable[[np.ndarray], Tuple[np.ndarray, np.ndarray]],
     init: np.ndarray,
@@ -207,6 +243,7 @@ def sample_nuts_batched(
     seed: Optional[int] = None,
     adapt_mass: bool = True,
     mass: str = "diag",
+    adaptation: str = "simple",
     max_depth: int = _MAX_DEPTH,
 ) -> Tuple[np.ndarray, dict]:
     """Run C lockstep NUTS chains over one batched evaluator.
@@ -223,6 +260,12 @@ def sample_nuts_batched(
         "dense" adapts a full covariance metric from the tuning window --
         required for strongly correlated posteriors (e.g. ODE parameters),
         where a diagonal metric mixes arbitrarily slowly.
+    adaptation : "simple" (default) or "windowed"
+        "simple" = one metric update at 60% of tune (matches
+        ``sample_nuts``, keeps C=1 bit-identity).  "windowed" = Stan-style
+        expanding windows (``metric_window_ends``): the metric is
+        re-estimated several times with step-size restarts, so early
+        burn-in never contaminates the final metric.
 
     Returns
     -------
@@ -259,6 +302,11 @@ def sample_nuts_batched(
     n_done = [0] * C
     windows: List[List[np.ndarray]] = [[] for _ in range(C)]
     mass_update_at = int(tune * 0.6)
+    if adaptation not in ("simple", "windowed"):
+        raise ValueError(f"unknown adaptation schedule: {adaptation!r}")
+    win_ends = metric_window_ends(tune) if adaptation == "windowed" else []
+    win_ptr = [0] * C
+    init_buffer = max(20, int(0.15 * tune))
     gens = [ch.transition() for ch in chains]
     pending: List[Optional[np.ndarray]] = [next(g) for g in gens]
     rounds = 0
@@ -272,22 +320,21 @@ def sample_nuts_batched(
         n_done[c] += 1
         if i < tune:
             ch.adapt_step_size()
-            if adapt_mass:
+            if adapt_mass and adaptation == "windowed":
+                if i >= init_buffer:
+                    windows[c].append(ch.q.copy())
+                if (win_ptr[c] < len(win_ends)
+                        and n_done[c] == win_ends[win_ptr[c]]):
+                    if len(windows[c]) > 10:
+                        _estimate_metric(ch, np.stack(windows[c]), mass)
+                        ch.reset_step_size_adaptation(0.25)
+                    windows[c].clear()
+                    win_ptr[c] += 1
+            elif adapt_mass:
                 windows[c].append(ch.q.copy())
                 if i == mass_update_at and len(windows[c]) > 10:
                     w = np.stack(windows[c][len(windows[c]) // 2:])
-                    if mass == "dense":
-                        # shrink toward the diagonal (Stan-style regularized
-                        # covariance metric) so the Cholesky stays stable
-                        n_w = w.shape[0]
-                        cov = np.atleast_2d(np.cov(w.T))
-                        lam = n_w / (n_w + 5.0)
-                        diag = np.diag(np.maximum(np.diag(cov), 1e-12))
-                        ch.set_dense_mass(lam * cov + (1 - lam) * diag
-                                          + 1e-12 * np.eye(cov.shape[0]))
-                    else:
-                        var = np.var(w, axis=0)
-                        ch.inv_mass = np.maximum(var, 1e-10)
+                    _estimate_metric(ch, w, mass)
                     windows[c].clear()
                     ch.reset_step_size_adaptation(0.25)
             if n_done[c] == tune:

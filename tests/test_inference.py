@@ -202,6 +202,45 @@ class TestNUTSBatched:
         ) > 300
         np.testing.assert_allclose(chain.mean(axis=(0, 2)), mu, atol=5e-4)
 
+    def test_windowed_adaptation_schedule(self):
+        """Expanding windows: step-size-only buffers at both ends, doubling
+        metric windows in between, remainder absorbed by the last window."""
+        from pytensor_federated_amd.inference.nuts_batched import metric_window_ends
+
+        ends = metric_window_ends(400)
+        assert ends == [85, 135, 360]
+        assert all(e2 > e1 for e1, e2 in zip(ends, ends[1:]))
+        # small tune still yields at least one update inside the buffers
+        assert metric_window_ends(80) == [55]
+        # term buffer is respected: last update leaves room to re-adapt eps
+        assert max(metric_window_ends(1000)) <= 1000 - 100
+
+    def test_windowed_adaptation_converges(self):
+        from pytensor_federated_amd.inference import sample_nuts_batched
+        from pytensor_federated_amd.inference.diagnostics import split_rhat
+
+        mu = np.array([0.8, 0.3, 0.6, 0.2])
+        sd = np.array([1e-3, 3e-4, 1e-3, 3e-4])
+        R = np.eye(4)
+        R[0, 1] = R[1, 0] = 0.99
+        R[2, 3] = R[3, 2] = 0.97
+        prec = np.linalg.inv(np.outer(sd, sd) * R)
+
+        def batched(theta):
+            d = theta - mu[:, None]
+            return -0.5 * np.einsum("kb,kj,jb->b", d, prec, d), -prec @ d
+
+        C = 8
+        init = np.tile(mu[:, None], (1, C)) * (
+            1 + 0.005 * np.random.RandomState(86).standard_normal((4, C))
+        )
+        chain, stats = sample_nuts_batched(
+            batched, init, draws=150, tune=300, step_size=5e-4, seed=88,
+            max_depth=8, mass="dense", adaptation="windowed",
+        )
+        assert max(split_rhat(chain[:, k, :].T) for k in range(4)) < 1.05
+        np.testing.assert_allclose(chain.mean(axis=(0, 2)), mu, atol=5e-4)
+
     def test_batched_linear_model_matches_conjugate(self):
         """Lockstep chains over a model's python-level batched evaluator
         reproduce the closed-form flat-prior posterior."""
