@@ -261,5 +261,8 @@ def sample_nuts(
                 # eps units changed with the metric: restart dual averaging
                 # at a conservative whitened-units step
                 sampler.reset_step_size_adaptation(0.25)
-    sampler.freeze_step_size()
+    if tune > 0:
+        # tune == 0 keeps the caller's step_size (freezing would reset it
+        # to exp(log_eps_bar) == 1.0 with no adaptation history)
+        sampler.freeze_step_size()
     return [sampler.step() for _ in range(draws)]

@@ -104,6 +104,22 @@ class TestNUTS:
         np.testing.assert_allclose(samples.mean(axis=0), mu, atol=5e-4)
         np.testing.assert_allclose(samples.std(axis=0), sd, rtol=0.3)
 
+    def test_tune_zero_keeps_user_step_size(self):
+        """tune=0 must not freeze: exp(log_eps_bar)=1.0 would silently
+        replace the caller's step size with no adaptation history."""
+        calls = [0]
+
+        def logp_grad(theta):
+            calls[0] += 1
+            t = np.asarray(theta, dtype=np.float64)
+            return np.asarray(-0.5 * t @ t), [-t]
+
+        chain = sample_nuts(
+            logp_grad, [np.zeros(2)], draws=20, tune=0, step_size=0.3, seed=24
+        )
+        assert len(chain) == 20
+        assert calls[0] > 20  # it actually sampled
+
     def test_divergences_rare_on_gaussian(self):
         from pytensor_federated_amd.inference.nuts import NUTS
 
