@@ -34,6 +34,7 @@
 
 #include <hip/hip_runtime.h>
 
+#include <atomic>
 #include <mutex>
 
 #include <string>
@@ -220,7 +221,7 @@ struct Worker {
     int* obs_dev = nullptr;       // int32[n_steps+1] step->obs row (-1 none)
     double* theta_dev = nullptr;  // f64[4]
     unsigned long long seq = 0;
-    int n_clients = 0;
+    std::atomic<int> n_clients{0};  // touched from every client thread
 };
 
 static unsigned short f32_to_bf16(float f) {
@@ -406,8 +407,8 @@ static void serve_client(Worker& w, int fd) {
         close(fd);
         return;
     }
-    w.n_clients++;
-    fprintf(stderr, "client connected (now %d)\n", w.n_clients);
+    int now = ++w.n_clients;
+    fprintf(stderr, "client connected (now %d)\n", now);
     std::vector<unsigned char> payload;
     while (true) {
         unsigned char hdr[5];
@@ -522,7 +523,7 @@ static void serve_client(Worker& w, int fd) {
             // carries GPU busy %, percent_ram carries VRAM use %.
             std::string out;
             put_varint(out, (1 << 3) | 0);  // n_clients
-            put_varint(out, (unsigned long long)w.n_clients);
+            put_varint(out, (unsigned long long)w.n_clients.load());
             float busy = -1.0f;
             for (int card = 0; card < 8 && busy < 0; ++card) {
                 char path[64];
@@ -555,8 +556,8 @@ static void serve_client(Worker& w, int fd) {
             write_frame(fd, 0xFF, "unknown frame type");
         }
     }
-    w.n_clients--;
-    fprintf(stderr, "client disconnected (now %d)\n", w.n_clients);
+    now = --w.n_clients;
+    fprintf(stderr, "client disconnected (now %d)\n", now);
     close(fd);
 }
 
