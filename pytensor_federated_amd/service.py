@@ -587,10 +587,6 @@ class ArraysToArraysServiceClient:
                 items.append(self._exporter.export(i))
             else:
                 items.append(ndarray_from_numpy(np.asarray(i)))
-        if self._exporter is not None and not any(
-            it.dtype.startswith("hipipc/") for it in items
-        ):
-            pass  # nothing exported this call
         input_arrays = InputArrays(items=items, uuid=str(uuid_module.uuid4()))
         cid = thread_pid_id(self)
         last_error: Optional[BaseException] = None
