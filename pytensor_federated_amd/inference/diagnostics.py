@@ -31,7 +31,6 @@ def split_rhat(samples) -> float:
     if n < 2:
         return float("nan")
     halves = np.concatenate([chains[:, :n], chains[:, n : 2 * n]], axis=0)
-    m = halves.shape[0]
     means = halves.mean(axis=1)
     w = halves.var(axis=1, ddof=1).mean()
     b = n * means.var(ddof=1)
@@ -80,9 +79,12 @@ def effective_sample_size(samples) -> float:
 
 
 def summary(chains_by_name: Dict[str, Sequence]) -> str:
-    """arviz-style text summary: mean, sd, 3%/97% HDI bounds, ESS, R-hat."""
+    """arviz-style text summary: mean, sd, central 94% interval, ESS, R-hat.
+
+    (Equal-tailed 3%/97% quantiles -- arviz reports an HDI; for the
+    near-symmetric posteriors these drivers target the two coincide.)"""
     rows = []
-    header = f"{'param':>12} {'mean':>10} {'sd':>10} {'hdi_3%':>10} {'hdi_97%':>10} {'ess':>8} {'r_hat':>6}"
+    header = f"{'param':>12} {'mean':>10} {'sd':>10} {'q3%':>10} {'q97%':>10} {'ess':>8} {'r_hat':>6}"
     rows.append(header)
     for name, samples in chains_by_name.items():
         arr = _to_chains(samples)
