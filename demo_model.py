@@ -15,7 +15,8 @@ import numpy as np
 import torch
 
 
-def run_model(host: str, ports, parallel: bool, map_steps: int, draws: int):
+def run_model(host: str, ports, parallel: bool, map_steps: int, draws: int,
+              sampler: str = "metropolis"):
     from pytensor_federated_amd.common import LogpGradServiceClient
     from pytensor_federated_amd.torch_ops import FederatedLogpGrad, LogpGradOp
 
@@ -41,6 +42,28 @@ def run_model(host: str, ports, parallel: bool, map_steps: int, draws: int):
         if step % 20 == 0:
             print(f"  MAP step {step}: logp={-float(loss):.3f} a={float(a):.4f} b={float(b):.4f}")
     print(f"MAP estimate: intercept={float(a):.4f} slope={float(b):.4f}")
+
+    if sampler == "nuts":
+        # ---- posterior: NUTS (each leapfrog = one fused federated call) --
+        from pytensor_federated_amd.inference import sample_nuts, summary
+
+        def logp_grad(theta):
+            ta = torch.tensor(theta[0], requires_grad=True, dtype=torch.float64)
+            tb = torch.tensor(theta[1], requires_grad=True, dtype=torch.float64)
+            lp = logp_op(ta, tb)
+            lp.backward()
+            return np.asarray(float(lp)), [
+                np.array([float(ta.grad), float(tb.grad)])
+            ]
+
+        draws_list = sample_nuts(
+            logp_grad, [np.array([float(a), float(b)])],
+            draws=draws, tune=max(draws // 2, 25), step_size=0.01, seed=0,
+        )
+        chain = np.stack([d[0] for d in draws_list])
+        print(f"NUTS posterior over {draws} draws:")
+        print(summary({"intercept": chain[:, 0], "slope": chain[:, 1]}))
+        return chain
 
     # ---- posterior: random-walk Metropolis ------------------------------
     rng = np.random.default_rng(0)
@@ -71,5 +94,7 @@ if __name__ == "__main__":
     parser.add_argument("--parallel", action="store_true")
     parser.add_argument("--map-steps", type=int, default=100)
     parser.add_argument("--draws", type=int, default=200)
+    parser.add_argument("--sampler", choices=["metropolis", "nuts"], default="metropolis")
     args, _ = parser.parse_known_args()
-    run_model(args.host, args.ports, args.parallel, args.map_steps, args.draws)
+    run_model(args.host, args.ports, args.parallel, args.map_steps, args.draws,
+              sampler=args.sampler)

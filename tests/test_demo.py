@@ -50,6 +50,11 @@ def test_demo_map_recovers_truth():
         # ground truth of generate_linear_dataset: intercept 1.5, slope 0.5
         assert abs(chain[:, 0].mean() - 1.5) < 0.3
         assert abs(chain[:, 1].mean() - 0.5) < 0.1
+        # NUTS over the same live workers (each leapfrog = one federated call)
+        chain = run_model("127.0.0.1", list(DEMO_PORTS), parallel=True,
+                          map_steps=0, draws=40, sampler="nuts")
+        assert abs(chain[:, 0].mean() - 1.5) < 0.3
+        assert abs(chain[:, 1].mean() - 0.5) < 0.1
     finally:
         for p in procs:
             if p.is_alive():
