@@ -40,8 +40,8 @@ def run_model(host: str, ports, parallel: bool, map_steps: int, draws: int,
         loss.backward()
         opt.step()
         if step % 20 == 0:
-            print(f"  MAP step {step}: logp={-float(loss):.3f} a={float(a):.4f} b={float(b):.4f}")
-    print(f"MAP estimate: intercept={float(a):.4f} slope={float(b):.4f}")
+            print(f"  MAP step {step}: logp={-loss.item():.3f} a={a.item():.4f} b={b.item():.4f}")
+    print(f"MAP estimate: intercept={a.item():.4f} slope={b.item():.4f}")
 
     if sampler == "nuts":
         # ---- posterior: NUTS (each leapfrog = one fused federated call) --
@@ -52,12 +52,12 @@ def run_model(host: str, ports, parallel: bool, map_steps: int, draws: int,
             tb = torch.tensor(theta[1], requires_grad=True, dtype=torch.float64)
             lp = logp_op(ta, tb)
             lp.backward()
-            return np.asarray(float(lp)), [
+            return np.asarray(lp.detach().item()), [
                 np.array([float(ta.grad), float(tb.grad)])
             ]
 
         draws_list = sample_nuts(
-            logp_grad, [np.array([float(a), float(b)])],
+            logp_grad, [np.array([a.item(), b.item()])],
             draws=draws, tune=max(draws // 2, 25), step_size=0.01, seed=0,
         )
         chain = np.stack([d[0] for d in draws_list])
@@ -67,7 +67,7 @@ def run_model(host: str, ports, parallel: bool, map_steps: int, draws: int,
 
     # ---- posterior: random-walk Metropolis ------------------------------
     rng = np.random.default_rng(0)
-    theta = np.array([float(a), float(b)])
+    theta = np.array([a.item(), b.item()])
     with torch.no_grad():
         cur_lp = float(logp_op(torch.tensor(theta[0]), torch.tensor(theta[1])))
     chain = []
