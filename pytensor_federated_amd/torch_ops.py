@@ -44,10 +44,12 @@ class _LogpGradFunction(torch.autograd.Function):
         if len(grads) != len(inputs):
             raise ValueError(f"Got {len(grads)} gradients for {len(inputs)} inputs.")
         ctx.grads = [
-            torch.as_tensor(np.asarray(g), dtype=t.dtype if t.is_floating_point() else torch.float64).to(t.device)
+            # copy: wire-decoded grads are zero-copy READ-ONLY views over
+            # the message bytes, which torch tensors must not alias
+            torch.tensor(np.asarray(g), dtype=t.dtype if t.is_floating_point() else torch.float64).to(t.device)
             for g, t in zip(grads, inputs)
         ]
-        return torch.as_tensor(np.asarray(logp), dtype=torch.float64)
+        return torch.as_tensor(float(np.asarray(logp)), dtype=torch.float64)
 
     @staticmethod
     def backward(ctx, g_logp):
