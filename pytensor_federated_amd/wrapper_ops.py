@@ -14,8 +14,10 @@ The fan-out machinery itself is framework-owned (``op_async.gather_evaluate``);
 these classes are thin shims, so the compute path (HIP kernels, RCCL shard
 sum) is identical whether driven from PyTensor, torch, or raw numpy.
 
-NOTE: untested in this repo's CI (pytensor is not installed in the ROCm
-image); the torch-native equivalents in ``torch_ops`` are the tested path.
+NOTE: pytensor itself is not installed in the ROCm image; the adapter
+logic is tested against a faithful graph-API stub
+(tests/test_wrapper_ops.py), and the torch-native equivalents in
+``torch_ops`` are the production-tested path.
 """
 from __future__ import annotations
 

@@ -1,0 +1,206 @@
+"""Minimal structural stand-in for the pytensor API surface wrapper_ops uses.
+
+pytensor is not installable in the ROCm image, so this stub lets the CPU
+suite exercise the ADAPTER'S OWN logic (make_node arity/typing, the
+ParallelAsyncOp input/output slicing, the dependence scan and fusion
+fixpoint, perform()'s output_storage layout) against faithful graph
+semantics: Variables know their owner Apply, toposort is a real
+dependency sort, and replace_all_validate really rewires consumers.
+
+What it deliberately does NOT emulate: symbolic math (grad graphs),
+shape/dtype inference, compilation.  Those paths stay reference-parity
+only and the torch-native equivalents in torch_ops.py are the tested
+production path.  Installed into sys.modules by tests/test_wrapper_ops.py
+in a SUBPROCESS so nothing leaks into other tests.
+"""
+import sys
+import types
+
+
+class TensorType:
+    def __init__(self, dtype="float64", shape=()):
+        self.dtype = dtype
+        self.shape = shape
+
+    def __call__(self):
+        return Variable(self)
+
+    def __eq__(self, other):
+        return isinstance(other, TensorType) and (self.dtype, self.shape) == (
+            other.dtype, other.shape)
+
+    def __hash__(self):
+        return hash((self.dtype, self.shape))
+
+
+class Variable:
+    def __init__(self, vtype, owner=None, index=0):
+        self.type = vtype
+        self.owner = owner
+        self.index = index
+
+    def __repr__(self):
+        return f"Var({self.type.dtype}, owner={id(self.owner) if self.owner else None})"
+
+
+class Apply:
+    def __init__(self, op, inputs, outputs):
+        self.op = op
+        self.inputs = list(inputs)
+        self.outputs = list(outputs)
+        for i, out in enumerate(self.outputs):
+            out.owner = self
+            out.index = i
+
+
+class Op:
+    def make_node(self, *inputs):
+        raise NotImplementedError
+
+    def __call__(self, *inputs):
+        node = self.make_node(*inputs)
+        return node.outputs if len(node.outputs) > 1 else node.outputs[0]
+
+
+class FromFunctionOp(Op):
+    def __init__(self, fn, itypes, otypes, infer_shape=None):
+        self._FromFunctionOp__fn = fn
+        self.itypes = itypes
+        self.otypes = otypes
+
+    def make_node(self, *inputs):
+        if len(inputs) != len(self.itypes):
+            raise ValueError(f"expected {len(self.itypes)} inputs, got {len(inputs)}")
+        return Apply(self, inputs, [t() for t in self.otypes])
+
+    def perform(self, node, inputs, output_storage, params=None):
+        outs = self._FromFunctionOp__fn(*inputs)
+        if not isinstance(outs, (list, tuple)):
+            outs = (outs,)
+        for i, out in enumerate(outs):
+            output_storage[i][0] = out
+
+
+def as_tensor_variable(x):
+    if isinstance(x, Variable):
+        return x
+    return Variable(TensorType("float64", ()))
+
+
+def dscalar():
+    return Variable(TensorType("float64", ()))
+
+
+class ReplaceValidate:
+    pass
+
+
+class GraphRewriter:
+    def add_requirements(self, fgraph):
+        pass
+
+    def apply(self, fgraph):
+        raise NotImplementedError
+
+    def rewrite(self, fgraph):
+        self.add_requirements(fgraph)
+        return self.apply(fgraph)
+
+
+class FunctionGraph:
+    def __init__(self, inputs, outputs):
+        self.inputs = list(inputs)
+        self.outputs = list(outputs)
+        self._features = []
+
+    def attach_feature(self, feature):
+        self._features.append(feature)
+
+    def _all_applies(self):
+        seen, order = set(), []
+
+        def visit(var):
+            app = var.owner
+            if app is None or id(app) in seen:
+                return
+            seen.add(id(app))
+            for inp in app.inputs:
+                visit(inp)
+            order.append(app)
+
+        for out in self.outputs:
+            visit(out)
+        return order
+
+    def toposort(self):
+        return self._all_applies()
+
+    def replace_all_validate(self, pairs, reason=None):
+        mapping = {id(old): new for old, new in pairs}
+        for app in self._all_applies():
+            app.inputs = [mapping.get(id(v), v) for v in app.inputs]
+        self.outputs = [mapping.get(id(v), v) for v in self.outputs]
+
+
+class _OptDB:
+    def __init__(self):
+        self.registered = {}
+
+    def register(self, name, rewriter, *tags, position=None):
+        if name in self.registered:
+            raise ValueError(f"rewrite {name} already registered")
+        self.registered[name] = (rewriter, tags, position)
+
+
+optdb = _OptDB()
+
+
+class DisconnectedType(TensorType):
+    pass
+
+
+def install():
+    """Insert the stub module tree as ``pytensor`` into sys.modules."""
+    pt = types.ModuleType("pytensor")
+    tensor = types.ModuleType("pytensor.tensor")
+    tensor.as_tensor_variable = as_tensor_variable
+    tensor.dscalar = dscalar
+    compile_mod = types.ModuleType("pytensor.compile")
+    compile_ops = types.ModuleType("pytensor.compile.ops")
+    compile_ops.FromFunctionOp = FromFunctionOp
+    compile_mode = types.ModuleType("pytensor.compile.mode")
+    compile_mode.optdb = optdb
+    graph = types.ModuleType("pytensor.graph")
+    graph_basic = types.ModuleType("pytensor.graph.basic")
+    graph_basic.Apply = Apply
+    graph_basic.Variable = Variable
+    graph_fg = types.ModuleType("pytensor.graph.fg")
+    graph_fg.FunctionGraph = FunctionGraph
+    graph_features = types.ModuleType("pytensor.graph.features")
+    graph_features.ReplaceValidate = ReplaceValidate
+    graph_op = types.ModuleType("pytensor.graph.op")
+    graph_op.Op = Op
+    graph_rewriting = types.ModuleType("pytensor.graph.rewriting")
+    graph_rewriting_basic = types.ModuleType("pytensor.graph.rewriting.basic")
+    graph_rewriting_basic.GraphRewriter = GraphRewriter
+    gradient = types.ModuleType("pytensor.gradient")
+    gradient.DisconnectedType = DisconnectedType
+    pt.tensor = tensor
+    pt.gradient = gradient
+    mods = {
+        "pytensor": pt,
+        "pytensor.tensor": tensor,
+        "pytensor.compile": compile_mod,
+        "pytensor.compile.ops": compile_ops,
+        "pytensor.compile.mode": compile_mode,
+        "pytensor.graph": graph,
+        "pytensor.graph.basic": graph_basic,
+        "pytensor.graph.fg": graph_fg,
+        "pytensor.graph.features": graph_features,
+        "pytensor.graph.op": graph_op,
+        "pytensor.graph.rewriting": graph_rewriting,
+        "pytensor.graph.rewriting.basic": graph_rewriting_basic,
+        "pytensor.gradient": gradient,
+    }
+    sys.modules.update(mods)
+    return mods
