@@ -1,4 +1,8 @@
-"""MAP estimation by Adam ascent on a LogpGradFunc."""
+"""MAP estimation by Adam ascent on a LogpGradFunc.
+
+The analog of the reference demo's ``pm.find_MAP`` step (reference
+demo_model.py:38) driven through this framework's LogpGradFunc contract.
+"""
 from __future__ import annotations
 
 from typing import List, Sequence, Tuple

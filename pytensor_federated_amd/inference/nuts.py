@@ -1,5 +1,8 @@
 """No-U-Turn Sampler (Hoffman & Gelman 2014, Algorithm 6) over a LogpGradFunc.
 
+The reference delegates posterior sampling to PyMC (reference
+demo_model.py:38-44 runs ``pm.sample``); this sampler provides that
+capability natively over the same fused-call contract.
 Each leapfrog step costs exactly ONE fused logp+grad call -- which on this
 framework is one HIP kernel + RCCL all-reduce (local engine) or one gRPC
 round trip (remote workers).  Dual-averaging step-size adaptation toward a

@@ -8,7 +8,9 @@ local Jacobian action; memory is O(n_steps * state), never O(n_steps *
 graph)).  The discrete adjoint is exact for the discretized system, so
 federated workers (each owning its private experiments batch) return
 gradients that sum exactly across shards -- same identity the linear/GLM
-models exploit.
+models exploit.  Like the reference's worker blackbox (reference
+demo_node.py:30-43), the model is served behind the ComputeFunc contract;
+only the compute inside changed.
 
 Batching: each worker integrates a whole batch ``u[B, D]`` of experiments
 in one vectorized sweep -- on an MI355X the RK4 right-hand side evaluates

@@ -7,8 +7,10 @@ One captured graph per rank holds the whole per-call pipeline:
     -> RCCL all_reduce([logp, ga, gb]) over xGMI          (multi-rank only)
     -> publish kernel (device epoch++ -> pinned mailbox {results, seq})
 
-Per evaluation the host writes two doubles into pinned memory, replays the
-graph, and spin-reads the mailbox seq -- no per-step torch dispatch, no
+This is the latency tier below the reference's hot path (its persistent
+EvaluateStream, reference service.py:150-158, still pays a Python+gRPC
+round trip per call): per evaluation the host writes two doubles into
+pinned memory, replays the graph, and spin-reads the mailbox seq -- no per-step torch dispatch, no
 stream sync call, no collective setup.  RCCL collectives are capturable on
 ROCm through torch's ProcessGroupNCCL, so the all-reduce rides inside the
 same replay.
