@@ -197,3 +197,42 @@ def test_fast_server_survives_malformed_frames(fast_servers):
     (out,) = client.evaluate(np.array(2.0), np.array(3.0))
     np.testing.assert_array_equal(out, np.array(6.0))
     del client
+
+
+@pytest.mark.timeout(180)
+def test_fast_server_fuzz_random_frames(fast_servers):
+    """Robustness: hundreds of random (type, length, payload) frames must
+    never kill the worker -- every reply is a well-formed frame (or the
+    connection closes), and real evaluations keep working afterwards."""
+    import asyncio as aio
+    import random
+
+    from pytensor_federated_amd.fastsock import MAGIC, _frame
+
+    rng = random.Random(91)
+
+    async def run():
+        for _ in range(40):
+            r, w = await aio.open_connection("127.0.0.1", FAST_PORTS[0])
+            w.write(MAGIC)
+            try:
+                for _ in range(rng.randint(1, 8)):
+                    ftype = rng.randrange(256)
+                    payload = bytes(rng.randrange(256) for _ in range(rng.randint(0, 64)))
+                    w.write(_frame(ftype, payload))
+                    await w.drain()
+                    hdr = await aio.wait_for(r.readexactly(5), timeout=10)
+                    length = int.from_bytes(hdr[1:5], "little")
+                    assert length < 1 << 20
+                    if length:
+                        await aio.wait_for(r.readexactly(length), timeout=10)
+            except (aio.IncompleteReadError, ConnectionError):
+                pass  # server may drop the connection; must not die
+            finally:
+                w.close()
+
+    asyncio.run(run())
+    client = ArraysToArraysServiceClient("127.0.0.1", FAST_PORTS[0], transport="fast")
+    (out,) = client.evaluate(np.array(3.0), np.array(4.0))
+    np.testing.assert_array_equal(out, np.array(12.0))
+    del client

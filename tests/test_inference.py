@@ -464,3 +464,37 @@ class TestDiagnostics:
                         "slope": rng.normal(0.5, 0.05, (2, 500))})
         assert "intercept" in text and "r_hat" in text
         assert len(text.splitlines()) == 3
+
+
+class TestLockstepEquivalenceProperty:
+    def test_bit_identity_over_random_posteriors(self):
+        """Property: for ANY 2-d Gaussian posterior and seed, C=1 lockstep
+        NUTS is draw-for-draw identical to the sequential sampler (stronger
+        form of the fixed-seed test above)."""
+        from pytensor_federated_amd.inference import sample_nuts_batched
+
+        rng = np.random.RandomState(31)
+        for trial in range(5):
+            mu = rng.standard_normal(2) * 2.0
+            a = rng.standard_normal((2, 2))
+            cov = a @ a.T + 0.3 * np.eye(2)
+            cov_inv = np.linalg.inv(cov)
+            single = gaussian_2d_logp_grad(mu, cov_inv)
+
+            def batched(theta, _single=single):
+                logp, (g,) = _single(theta[:, 0])
+                return np.array([float(logp)]), np.asarray(g).reshape(2, 1)
+
+            seed = int(rng.randint(0, 10_000))
+            tune = int(rng.randint(10, 60))
+            draws = int(rng.randint(10, 50))
+            init = rng.standard_normal(2)
+            ref = np.stack([
+                d[0] for d in sample_nuts(single, [init.copy()],
+                                          draws=draws, tune=tune, seed=seed)
+            ])
+            chain, _ = sample_nuts_batched(
+                batched, init[:, None].copy(), draws=draws, tune=tune, seed=seed
+            )
+            np.testing.assert_array_equal(chain[:, :, 0], ref,
+                                          err_msg=f"trial {trial}")
