@@ -1,6 +1,11 @@
 #!/usr/bin/env python3
 """Flagship benchmark: federated logp+grad calls/sec (BASELINE.json metric).
 
+The JSON line's ``metric`` is "logp+grad calls/sec (whole node)" -- the
+BASELINE.json metric string minus its trailing ", linear-regression demo
+at 1/2/4/8 MI355X" clause, which describes the driver-run 1/2/4/8-GPU
+scaling sweep rather than a single run; ``config`` names the model.
+
 Measures the linear-regression demo config on 1..8 MI355X GPUs:
 each rank (GPU) owns a private synthetic shard of 1e7 bf16 rows; one
 "step" = one full federated logp+grad evaluation -- broadcast theta,
