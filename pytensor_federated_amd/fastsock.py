@@ -37,14 +37,34 @@ T_EVAL_R = 0x81
 T_LOAD_R = 0x82
 T_ERR = 0xFF
 
+#: Upper bound on one frame's payload (default 256 MiB).  The length header
+#: is an untrusted 32-bit value; without this cap a single hostile frame
+#: forces a ~4 GiB allocation in ``readexactly``.  Override via
+#: ``FED_FASTSOCK_MAX_FRAME`` (bytes) for genuinely larger arrays.
+import os as _os
+
+MAX_FRAME_BYTES = int(_os.environ.get("FED_FASTSOCK_MAX_FRAME", 256 * 1024 * 1024))
+
+
+class FrameTooLargeError(ConnectionError):
+    """A frame header announced a payload above :data:`MAX_FRAME_BYTES`."""
+
 
 def _frame(frame_type: int, payload: bytes) -> bytes:
+    if len(payload) > MAX_FRAME_BYTES:
+        raise FrameTooLargeError(
+            f"refusing to send {len(payload)}-byte frame (cap {MAX_FRAME_BYTES})"
+        )
     return bytes([frame_type]) + len(payload).to_bytes(4, "little") + payload
 
 
 async def _read_frame(reader: asyncio.StreamReader):
     hdr = await reader.readexactly(5)
     length = int.from_bytes(hdr[1:5], "little")
+    if length > MAX_FRAME_BYTES:
+        raise FrameTooLargeError(
+            f"peer announced {length}-byte frame (cap {MAX_FRAME_BYTES})"
+        )
     payload = await reader.readexactly(length) if length else b""
     return hdr[0], payload
 
@@ -63,12 +83,14 @@ async def start_fast_server_async(service, bind: str, port: int):
             return
         service._n_clients += 1
         _log.info("fast client connected. Now serving %i clients.", service._n_clients)
+        # Per-connection export region (requests on one connection are
+        # sequential, so reset-between-requests is race-free here).
+        exporter = service._new_exporter()
         try:
             while True:
                 ftype, payload = await _read_frame(reader)
                 if ftype == T_EVAL:
                     try:
-                        exporter = getattr(service, "_exporter", None)
                         if exporter is not None:
                             exporter.reset()
                         out = _run_compute_func(

@@ -62,10 +62,13 @@ def effective_sample_size(samples) -> float:
     acov = np.fft.irfft(f * np.conj(f), n=n_fft, axis=1)[:, :n].real
     acov /= n
     rho = 1.0 - (w - acov.mean(axis=0)) / var_plus  # combined autocorrelation
-    # Geyer: sum consecutive pairs while positive, enforce monotone decrease
-    tau = 1.0
+    # Geyer initial-monotone estimator, Stan/arviz pairing: sum pairs
+    # (rho[0]+rho[1]), (rho[2]+rho[3]), ... while positive, with rho[0] the
+    # computed lag-0 value (slightly below 1 for finite n), enforce monotone
+    # non-increase, then tau = -1 + 2*sum(pairs).
+    tau = -1.0
     prev_pair = float("inf")
-    t = 1
+    t = 0
     while t + 1 < n:
         pair = rho[t] + rho[t + 1]
         if pair < 0:

@@ -131,18 +131,59 @@ class DeviceArrayExporter:
         return nda
 
 
-#: importer-side cache: handle bytes -> mapped base pointer
+#: importer-side cache: handle bytes -> mapped base pointer.  Bounded: when
+#: it exceeds _OPENED_MAX distinct exporter regions, the least recently used
+#: mapping is closed (after a device sync so no in-flight copy still reads it).
 _opened: Dict[bytes, int] = {}
+_OPENED_MAX = 64
+
+
+def _evict_opened_lru() -> None:
+    import torch
+
+    lib = _lib()
+    while len(_opened) > _OPENED_MAX:
+        handle, base = next(iter(_opened.items()))
+        torch.cuda.synchronize()
+        lib.fed_ipc_close(base)
+        del _opened[handle]
 
 
 def device_ndarray_to_torch(nda: Ndarray, device=None):
-    """Map the exporter's region (cached) and D2D-copy into a fresh tensor."""
+    """Map the exporter's region (cached) and D2D-copy into a fresh tensor.
+
+    The wire header is untrusted: ``nbytes`` must equal the byte size implied
+    by ``shape``/``dtype`` and ``offset`` must be non-negative, or the message
+    is rejected before any device copy happens.
+    """
     import torch
 
     if not is_device_ndarray(nda):
         raise ValueError("not a device ndarray")
-    lib = _lib()
+    if len(nda.data) != _HEADER.size:
+        raise ValueError(
+            f"device ndarray header must be {_HEADER.size} bytes, got {len(nda.data)}"
+        )
     handle, offset, nbytes = _HEADER.unpack(bytes(nda.data))
+    dtype_name = nda.dtype[len(DEVICE_DTYPE_PREFIX) :]
+    dtype = getattr(torch, dtype_name, None)
+    if not isinstance(dtype, torch.dtype):
+        raise ValueError(f"unknown device ndarray dtype {dtype_name!r}")
+    itemsize = torch.empty(0, dtype=dtype).element_size()
+    numel = 1
+    for s in nda.shape:
+        if s < 0:
+            raise ValueError(f"negative dimension in shape {nda.shape}")
+        numel *= s
+    expected = numel * itemsize
+    if nbytes != expected:
+        raise ValueError(
+            f"device ndarray nbytes={nbytes} does not match "
+            f"shape {list(nda.shape)} x {dtype} = {expected} bytes"
+        )
+    if offset < 0:
+        raise ValueError(f"negative device ndarray offset {offset}")
+    lib = _lib()
     base = _opened.get(handle)
     if base is None:
         ptr = ctypes.c_void_p()
@@ -151,7 +192,11 @@ def device_ndarray_to_torch(nda: Ndarray, device=None):
             raise RuntimeError(f"hipIpcOpenMemHandle failed ({rc})")
         base = ptr.value
         _opened[handle] = base
-    dtype = getattr(torch, nda.dtype[len(DEVICE_DTYPE_PREFIX) :])
+        _evict_opened_lru()
+    else:
+        # refresh LRU position
+        del _opened[handle]
+        _opened[handle] = base
     if device is None:
         device = torch.device("cuda", torch.cuda.current_device())
     out = torch.empty(nda.shape, dtype=dtype, device=device)

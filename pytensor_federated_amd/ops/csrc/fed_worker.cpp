@@ -409,12 +409,23 @@ static void serve_client(Worker& w, int fd) {
     }
     int now = ++w.n_clients;
     fprintf(stderr, "client connected (now %d)\n", now);
+    // Frame cap: the length header is untrusted; without a bound one hostile
+    // frame forces a ~4 GiB allocation (mirrors fastsock.MAX_FRAME_BYTES).
+    unsigned long max_frame = 256ul * 1024 * 1024;
+    if (const char* mf = getenv("FED_FASTSOCK_MAX_FRAME")) {
+        unsigned long v = strtoul(mf, nullptr, 10);
+        if (v > 0) max_frame = v;
+    }
     std::vector<unsigned char> payload;
     while (true) {
         unsigned char hdr[5];
         if (!read_exact(fd, hdr, 5)) break;
         unsigned int ln;
         memcpy(&ln, hdr + 1, 4);
+        if ((unsigned long)ln > max_frame) {
+            fprintf(stderr, "frame of %u bytes exceeds cap %lu; closing client\n", ln, max_frame);
+            break;
+        }
         payload.resize(ln);
         if (ln && !read_exact(fd, payload.data(), ln)) break;
         static std::mutex eval_mu;

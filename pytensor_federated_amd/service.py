@@ -125,13 +125,30 @@ class ArraysToArraysService:
         self._compute_func = compute_func
         self._n_clients = 0
         self._report_gpu_load = report_gpu_load
-        self._exporter = None
-        if device_arrays:
-            from .npproto.device import DeviceArrayExporter
-
-            self._exporter = DeviceArrayExporter()
+        self._device_arrays = device_arrays
+        # Unary device-array replies rotate through a small ring of export
+        # regions so a region is only reused after _UNARY_RING-1 intervening
+        # requests (the protocol has no consumption ack; see evaluate()).
+        self._unary_exporters = None
+        self._unary_idx = 0
         # Prime psutil's CPU monitoring so the first GetLoad is meaningful.
         psutil.getloadavg()
+
+    _UNARY_RING = 4
+
+    def _new_exporter(self):
+        """Fresh device-array exporter, or None when device_arrays is off.
+
+        Streamed connections each own one exporter: requests on one stream
+        are strictly sequential and the client D2D-copies reply k before
+        sending k+1, so resetting between requests is race-free *per
+        connection* (a single shared exporter was not -- two concurrent
+        streams could overwrite each other's regions mid-read)."""
+        if not self._device_arrays:
+            return None
+        from .npproto.device import DeviceArrayExporter
+
+        return DeviceArrayExporter()
 
     # -- telemetry ----------------------------------------------------
     @property
@@ -156,18 +173,26 @@ class ArraysToArraysService:
 
     # -- RPC handlers (grpc.aio behavior functions) --------------------
     async def evaluate(self, input_arrays: InputArrays, context=None) -> OutputArrays:
-        if self._exporter is not None:
-            self._exporter.reset()  # previous reply was consumed (1 in flight)
-        return _run_compute_func(input_arrays, self._compute_func, self._exporter)
+        exporter = None
+        if self._device_arrays:
+            if self._unary_exporters is None:
+                self._unary_exporters = [
+                    self._new_exporter() for _ in range(self._UNARY_RING)
+                ]
+            exporter = self._unary_exporters[self._unary_idx % self._UNARY_RING]
+            self._unary_idx += 1
+            exporter.reset()
+        return _run_compute_func(input_arrays, self._compute_func, exporter)
 
     async def evaluate_stream(self, request_iterator, context=None):
+        exporter = self._new_exporter()  # per-connection; see _new_exporter
         self._n_clients += 1
         _log.info("A client started a stream. Now serving %i clients.", self._n_clients)
         try:
             async for input_arrays in request_iterator:
-                if self._exporter is not None:
-                    self._exporter.reset()
-                yield _run_compute_func(input_arrays, self._compute_func, self._exporter)
+                if exporter is not None:
+                    exporter.reset()
+                yield _run_compute_func(input_arrays, self._compute_func, exporter)
         finally:
             self._n_clients -= 1
             _log.info("A client ended a stream. Now serving %i clients.", self._n_clients)
@@ -380,8 +405,35 @@ class ClientPrivates:
         return await ClientPrivates.connect(host, port, transport=transport)
 
 
-#: module-global connection cache; see :class:`ClientPrivates`.
+#: module-global connection cache; see :class:`ClientPrivates`.  Bounded:
+#: entries beyond _PRIVATES_MAX are evicted oldest-first with their channel
+#: closed, so long-lived processes churning clients/threads cannot leak
+#: connections (the common cids -- one per live (client, pid, thread) --
+#: stay far below the cap).
 _privates: Dict[str, ClientPrivates] = {}
+_PRIVATES_MAX = 256
+
+
+async def _evict_privates_lru() -> None:
+    while len(_privates) > _PRIVATES_MAX:
+        cid, privates = next(iter(_privates.items()))
+        del _privates[cid]
+        try:
+            if privates.stream is not None:
+                privates.stream.cancel()
+            if privates.channel is not None:
+                await privates.channel.close()
+        except Exception:
+            pass
+        _log.info("Evicted idle connection %s (cache > %i).", cid, _PRIVATES_MAX)
+        _client_exporters.pop(cid, None)
+
+
+#: client-side device-array exporters, keyed like ``_privates`` so each
+#: (client, pid, thread) owns its region (an exporter on the client object
+#: itself would be shared across threads -- an overwrite race -- and would
+#: break pickling into multiprocessing workers).
+_client_exporters: Dict[str, object] = {}
 
 
 def _after_fork_in_child() -> None:
@@ -407,10 +459,18 @@ def thread_pid_id(obj: object) -> str:
 
 
 async def _streamed_evaluate(stream, input_arrays: InputArrays) -> OutputArrays:
-    """One send + one receive on the persistent stream (the hot path)."""
+    """One send + one receive on the persistent stream (the hot path).
+
+    Stream end is detected structurally: grpc.aio's ``read()`` returns the
+    ``grpc.aio.EOF`` sentinel (checked by identity) once the server closes;
+    anything that is not a deserialized ``OutputArrays`` message likewise
+    means the stream no longer carries replies.
+    """
+    import grpc.aio
+
     await stream.write(input_arrays)
     output = await stream.read()
-    if output is None or type(output).__name__ == "EOF" or repr(output) == "EOF":
+    if output is None or output is grpc.aio.EOF or not isinstance(output, OutputArrays):
         raise ConnectionError("Bidirectional stream was closed by the server.")
     return output
 
@@ -436,6 +496,7 @@ async def _connect_evaluate_async(
                 client._host, client._port, transport=client._transport
             )
         _privates[cid] = privates
+        await _evict_privates_lru()
 
     if client._transport == "fast":
         # the persistent fast connection IS the stream; unary == stream
@@ -519,10 +580,10 @@ class ArraysToArraysServiceClient:
         self._retries = retries
         self._transport = transport
         self._device_arrays = device_arrays
-        self._exporter = None
 
     def __del__(self):
         cid = thread_pid_id(self)
+        _client_exporters.pop(cid, None)
         privates = _privates.pop(cid, None)
         if privates is None:
             return
@@ -573,6 +634,8 @@ class ArraysToArraysServiceClient:
         if retries is None:
             retries = self._retries
 
+        cid = thread_pid_id(self)
+        exporter = _client_exporters.get(cid)
         items = []
         for i in inputs:
             if (
@@ -580,15 +643,15 @@ class ArraysToArraysServiceClient:
                 and type(i).__module__ == "torch"
                 and getattr(i, "is_cuda", False)
             ):
-                if self._exporter is None:
+                if exporter is None:
                     from .npproto.device import DeviceArrayExporter
 
-                    self._exporter = DeviceArrayExporter()
-                items.append(self._exporter.export(i))
+                    exporter = DeviceArrayExporter()
+                    _client_exporters[cid] = exporter
+                items.append(exporter.export(i))
             else:
                 items.append(ndarray_from_numpy(np.asarray(i)))
         input_arrays = InputArrays(items=items, uuid=str(uuid_module.uuid4()))
-        cid = thread_pid_id(self)
         last_error: Optional[BaseException] = None
         for attempt in range(retries + 1):
             try:
@@ -601,8 +664,8 @@ class ArraysToArraysServiceClient:
                         decoded.append(device_ndarray_to_torch(item))
                     else:
                         decoded.append(ndarray_to_numpy(item))
-                if self._exporter is not None:
-                    self._exporter.reset()  # server consumed the request arrays
+                if exporter is not None:
+                    exporter.reset()  # server consumed the request arrays
                 return decoded
             except (grpc.RpcError, ConnectionError, OSError) as ex:
                 last_error = ex

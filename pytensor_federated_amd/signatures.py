@@ -12,19 +12,19 @@ import numpy as np
 __all__ = ["ComputeFunc", "LogpFunc", "LogpGradFunc"]
 
 ComputeFunc = Callable[
-    [Sequence[np.ndarray]],  # arbitrary number of input arrays
-    Sequence[np.ndarray],  # arbitrary number of output arrays
+    [Sequence[np.ndarray]],  # any N parameter/data arrays
+    Sequence[np.ndarray],  # any M result arrays
 ]
 """Generic compute function: multiple arrays in, multiple arrays out."""
 
 LogpFunc = Callable[
-    [Sequence[np.ndarray]],  # arbitrary number of input arrays
-    np.ndarray,  # scalar log-p
+    [Sequence[np.ndarray]],  # any N parameter arrays
+    np.ndarray,  # 0-d log-probability
 ]
 """Log-probability function without gradients (e.g. a log-likelihood)."""
 
 LogpGradFunc = Callable[
-    [Sequence[np.ndarray]],  # arbitrary number of input arrays
-    Tuple[np.ndarray, Sequence[np.ndarray]],  # scalar log-p, grads w.r.t. each input
+    [Sequence[np.ndarray]],  # any N parameter arrays
+    Tuple[np.ndarray, Sequence[np.ndarray]],  # (0-d log-p, one grad per input)
 ]
 """Log-probability function with gradients w.r.t. its inputs."""

@@ -3,9 +3,7 @@ from __future__ import annotations
 
 import asyncio
 import logging
-from typing import Callable, Iterable, List, Optional, TypeVar
-
-import numpy as np
+from typing import Callable, Iterable, Optional, TypeVar
 
 T = TypeVar("T")
 _log = logging.getLogger(__file__)
@@ -20,13 +18,18 @@ def argmin_none_or_func(
     """Argmin of ``func`` over non-``None`` items; ``None`` if all are ``None``.
 
     Used by the balanced-connect logic to pick the least-loaded live worker
-    while ignoring dead ones (reference utils.py:13-34).
+    while ignoring dead ones (reference utils.py:13-34).  Ties resolve to the
+    earliest index, matching the shuffled-server-list semantics.
     """
-    items = list(items)
-    if not any(item is not None for item in items):
-        return None
-    values: List[float] = [np.inf if item is None else func(item) for item in items]
-    return int(np.argmin(values))
+    best_idx: Optional[int] = None
+    best_val: Optional[float] = None
+    for i, item in enumerate(items):
+        if item is None:
+            continue
+        val = float(func(item))
+        if best_val is None or val < best_val:
+            best_idx, best_val = i, val
+    return best_idx
 
 
 def get_useful_event_loop() -> asyncio.AbstractEventLoop:
@@ -44,7 +47,7 @@ def get_useful_event_loop() -> asyncio.AbstractEventLoop:
         if not hasattr(loop, "_nest_patched"):
             import nest_asyncio
 
-            _log.debug("Event loop is already running. Patching with nest_asyncio.")
+            _log.debug("A loop is running here; applying nest_asyncio re-entrance patch.")
             nest_asyncio.apply(loop)
         return loop
     try:
