@@ -224,6 +224,32 @@ def main():
             return host_buf
         return None
 
+    # Per-call collective cost (multi-rank only): mean wall time of one
+    # eager all_reduce of the 3..K-double [logp, grads] buffer, measured
+    # outside the bench loop so SCALE runs carry the latency diagnostic
+    # whichever path (graphed or engine) the steps take.
+    allreduce_us = None
+    if distributed:
+        import torch.distributed as dist
+
+        probe = torch.zeros(
+            3 if args.model == "linear" else int(np.prod(theta0.shape)) + 1,
+            dtype=torch.float64,
+            device=device,
+        )
+        for _ in range(20):  # warm the communicator/algorithm choice
+            dist.all_reduce(probe)
+        if have_gpu:
+            torch.cuda.synchronize()
+        dist.barrier()
+        tar0 = time.perf_counter()
+        reps = 200
+        for _ in range(reps):
+            dist.all_reduce(probe)
+        if have_gpu:
+            torch.cuda.synchronize()
+        allreduce_us = (time.perf_counter() - tar0) / reps * 1e6
+
     def barrier_sync():
         if distributed:
             import torch.distributed as dist
@@ -270,9 +296,18 @@ def main():
             "data": "synthetic",
             "config": {
                 "model": config_model,
-                "parallelism": f"federated shard-dp x{n_gpus} (RCCL all-reduce)",
-                "per_step": "theta -> fused logp+grad kernel per shard -> "
-                "all_reduce([logp,*grads]) -> host readback",
+                "parallelism": (
+                    f"federated shard-dp x{n_gpus} (RCCL all-reduce)"
+                    if n_gpus > 1
+                    else "federated shard x1 (single shard, no collective)"
+                ),
+                "allreduce_us": allreduce_us,
+                "per_step": (
+                    "theta -> fused logp+grad kernel per shard -> "
+                    "all_reduce([logp,*grads]) -> host readback"
+                    if n_gpus > 1
+                    else "theta -> fused logp+grad kernel -> host readback"
+                ),
                 "readback": readback,
                 "device": str(device),
                 "kernels": bool(use_kernels is None),

@@ -119,3 +119,60 @@ class TestMultiShardDispatcher:
     def test_empty_raises(self):
         with pytest.raises(ValueError):
             MultiShardDispatcher([])
+
+
+def _agreed_rank_entry(rank, world_size, port, result_queue):
+    """Rank entry: attempt graphed create_agreed on CPU (capture must fail on
+    every rank) and report that the agreed protocol returns None without
+    deadlocking the collective sequence."""
+    import torch.distributed as dist
+
+    from pytensor_federated_amd.parallel.graphed import GraphedLinearEngine
+
+    dist.init_process_group(
+        backend="gloo",
+        init_method=f"tcp://127.0.0.1:{port}",
+        rank=rank,
+        world_size=world_size,
+    )
+    try:
+        x, y = generate_linear_dataset(64, seed=3)
+        s = shard_slice(64, rank, world_size)
+        model = GaussianLinearModel(x[s], y[s], sigma=0.4)
+        engine = GraphedLinearEngine.create_agreed(model, distributed=True)
+        # after the agreed outcome the communicator must still be usable
+        t = torch.ones(1, dtype=torch.float64)
+        dist.all_reduce(t)
+        result_queue.put((rank, engine is None, float(t[0])))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_create_agreed_degrades_uniformly_across_ranks():
+    """Multi-rank de-risk for the hipgraph path: when capture fails (here: no
+    GPU on any rank), EVERY rank gets None back and the process group remains
+    usable -- the property that keeps an 8-GPU job from deadlocking when one
+    rank's capture fails."""
+    world_size = 2
+    port = 29572
+    ctx = multiprocessing.get_context("spawn")
+    queue = ctx.Queue()
+    procs = [
+        ctx.Process(
+            target=_agreed_rank_entry, args=(r, world_size, port, queue), daemon=True
+        )
+        for r in range(world_size)
+    ]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world_size):
+        rank, engine_is_none, reduced = queue.get(timeout=240)
+        results[rank] = (engine_is_none, reduced)
+    for p in procs:
+        p.join(timeout=30)
+    for rank in range(world_size):
+        engine_is_none, reduced = results[rank]
+        assert engine_is_none  # uniform fallback on every rank
+        assert reduced == world_size  # communicator still aligned afterwards
