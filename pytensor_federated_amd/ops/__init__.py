@@ -397,7 +397,7 @@ def logistic_glm_logp_grad_batched(
     theta_t = theta.detach().t().contiguous().to(device=X.device, dtype=torch.bfloat16)
     if out is None:
         out = torch.empty(B + K * B, dtype=torch.float64, device=X.device)
-    ws = _workspace(X.device, f"logistic_batched{K}", 304 * (B + K * B), dtype=torch.float32)
+    ws = _workspace(X.device, f"logistic_batched{K}", 768 * (B + K * B), dtype=torch.float32)
     rc = lib.fed_logistic_glm_batched(
         X.data_ptr(), y.data_ptr(), n, K,
         theta_t.data_ptr(), out.data_ptr(), ws.data_ptr(), ws.numel() * 4,

@@ -1150,6 +1150,199 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched(
 
 
 // ---------------------------------------------------------------------------
+// Batched logistic v2: fewer barriers, swizzled LDS, write-after-barrier
+// ---------------------------------------------------------------------------
+// Round-1 PMC on v1 (profiles/PROFILES.md): 60% wave-park cycles (staging
+// written BEFORE each chunk barrier, loads only get the MFMA span) and
+// 10.8% LDS bank-conflict cycles (phase-B transposed u16 reads: rows r/r+8
+// land on one bank at the 16B-aligned row stride).  v2 changes:
+//
+//  * write-after-barrier staging (T14): the LDS write of chunk c+1 happens
+//    AFTER barrier c and chunk c+2's global loads re-issue immediately, so
+//    every wave gives its loads a full chunk-step of latency instead of
+//    only the MFMA span.
+//  * XOR swizzle of the 16-byte group index, key = ((row>>3)&1)<<1 --
+//    rows r and r+8 (the colliding pair within a 32-lane half) get their
+//    group-bit-1 flipped against each other, separating their banks for
+//    the phase-B scalar reads while leaving the b128 write/read patterns
+//    near-conflict-free.
+//  * Theta fragments read straight from global (32 KB, L2-resident),
+//    freeing 33 KB LDS: the block's footprint drops to ~38 KB, so THREE
+//    blocks fit per CU (12 waves in flight vs v1's 8).
+//  * out-of-range rows handled by address clamping (their logp/resid
+//    contributions are masked later) instead of per-element zero-fill
+//    branches, which de-pipeline hipcc's load scheduling.
+
+template <int K>
+__global__ __launch_bounds__(256, 3) void k_logistic_glm_batched_v2(
+    const unsigned short* __restrict__ X,   // [N][K] bf16
+    const unsigned short* __restrict__ y,   // [N] bf16
+    long long n_rows,
+    const unsigned short* __restrict__ theta_t,  // [BCH][K] bf16 (transposed)
+    float* __restrict__ slab                     // [grid][BCH + K*BCH]
+) {
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    constexpr int n_chunks = K / BL_CHUNK;
+
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    unsigned short* x_lds = (unsigned short*)smem;              // [2][BL_ROWS][BL_CHUNK+XPAD]
+    const int x_stride = BL_CHUNK + XPAD;
+    const int x_buf = BL_ROWS * x_stride;
+    unsigned short* rt_lds = x_lds + 2 * x_buf;                 // [BCH][BL_ROWS+RPAD]
+    const int rt_stride = BL_ROWS + RPAD;
+    float* y_lds = (float*)(rt_lds + BCH * rt_stride + 8);      // [BL_ROWS]
+    float* red_lds = y_lds + BL_ROWS;                           // [256]
+
+    typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
+    typedef __attribute__((ext_vector_type(4))) float f32x4_t;
+    union frag_u { bf16x8_t v; unsigned short u[8]; U4 q; };
+
+    f32x4_t g_acc[n_chunks * 2];
+#pragma unroll
+    for (int t = 0; t < n_chunks * 2; ++t) g_acc[t] = (f32x4_t){0.f, 0.f, 0.f, 0.f};
+    float logp_acc = 0.f;  // this lane's chain partial (chain = lane&15)
+
+    // staging: thread owns rows {r0, r0+16, r0+32, r0+48}, one 16B group each
+    const int st_r0 = threadIdx.x / 16;               // 0..15
+    const int st_g = threadIdx.x % 16;                // 16B group within chunk
+    const int st_key = ((st_r0 >> 3) & 1) << 1;       // swizzle key (rows +16 share it)
+    const long long row_max = n_rows - 1;
+
+    const long long n_tiles = (n_rows + BL_ROWS - 1) / BL_ROWS;
+    for (long long tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
+        const long long row0 = tile * BL_ROWS;
+        if (threadIdx.x < BL_ROWS) {
+            const long long r = row0 + threadIdx.x;
+            y_lds[threadIdx.x] = r < n_rows ? bf16_bits_to_f32(y[r]) : 0.f;
+        }
+
+        U4 ld[4];
+        // global loads are linear (coalesced); the XOR swizzle is applied at
+        // the LDS WRITE, and compensated at both read sites.
+#define LOAD_CHUNK2(c)                                                             _Pragma("unroll") for (int rr = 0; rr < 4; ++rr) {                                 long long row = row0 + st_r0 + rr * 16;                                        row = row > row_max ? row_max : row;                                           ld[rr] = *(const U4*)&X[row * (long long)K + (c) * BL_CHUNK + st_g * 8];   }
+#define WRITE_CHUNK2(buf)                                                          _Pragma("unroll") for (int rr = 0; rr < 4; ++rr)                                   *(U4*)&x_lds[(buf) * x_buf + (st_r0 + rr * 16) * x_stride +                                 (st_g ^ st_key) * 8] = ld[rr];
+
+        LOAD_CHUNK2(0)
+        WRITE_CHUNK2(0)
+        LOAD_CHUNK2(1)
+        int cur = 0;
+
+        // ---- phase A: Z = X . Theta (theta fragments from L2) ----
+        // (not unrolled: full unrolling lets the scheduler hoist every
+        // chunk's loads and blow past the 2-wave/SIMD register budget;
+        // phase B must unroll for static g_acc indexing, this loop not)
+        f32x4_t z_acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll 1
+        for (int c = 0; c < n_chunks; ++c) {
+            __syncthreads();  // buf[cur] visible (and y_lds on c==0)
+            if (c + 1 < n_chunks) {
+                WRITE_CHUNK2(cur ^ 1)
+                if (c + 2 < n_chunks) LOAD_CHUNK2(c + 2)
+            }
+            const int arow = wid * 16 + (lane & 15);
+            const int akey = (((arow >> 3) & 1) << 1);
+#pragma unroll
+            for (int ks = 0; ks < BL_CHUNK / 32; ++ks) {
+                frag_u a, b;
+                const int ag = (ks * 4 + (lane >> 4)) ^ akey;
+                a.q = *(U4*)&x_lds[cur * x_buf + arow * x_stride + ag * 8];
+                const int bk = c * BL_CHUNK + ks * 32 + (lane >> 4) * 8;
+                b.q = *(const U4*)&theta_t[(lane & 15) * K + bk];
+                z_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, z_acc, 0, 0, 0);
+            }
+            cur ^= 1;
+        }
+
+        // ---- logp + R from Z (prefetch chunk 0 for phase B) ----
+        LOAD_CHUNK2(0)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int row_in_wave = (lane >> 4) * 4 + r;
+            const int row_in_tile = wid * 16 + row_in_wave;
+            const long long row = row0 + row_in_tile;
+            const int chain = lane & 15;
+            float z = z_acc[r];
+            float yv = y_lds[row_in_tile];
+            float resid = 0.f;
+            if (row < n_rows) {
+                const float sp = fmaxf(z, 0.f) + log1pf(__expf(-fabsf(z)));
+                logp_acc += yv * z - sp;
+                resid = yv - 1.f / (1.f + __expf(-z));
+            }
+            union { float f; unsigned int u; } cv;
+            cv.f = resid;
+            const unsigned int rnd = 0x7fff + ((cv.u >> 16) & 1);
+            rt_lds[chain * rt_stride + row_in_tile] = (unsigned short)((cv.u + rnd) >> 16);
+        }
+        __syncthreads();  // R complete; x_lds free
+        WRITE_CHUNK2(0)
+        LOAD_CHUNK2(1)
+        cur = 0;
+
+        // ---- phase B: G += X_chunk^T . R (chunks L2-hot) ----
+#pragma unroll
+        for (int c = 0; c < n_chunks; ++c) {
+            __syncthreads();
+            if (c + 1 < n_chunks) {
+                WRITE_CHUNK2(cur ^ 1)
+                if (c + 2 < n_chunks) LOAD_CHUNK2(c + 2)
+            }
+#pragma unroll
+            for (int t2 = 0; t2 < 2; ++t2) {
+                const int kcol0 = wid * 32 + t2 * 16;
+                f32x4_t acc = g_acc[c * 2 + t2];
+#pragma unroll
+                for (int rs = 0; rs < 2; ++rs) {
+                    frag_u a, b;
+                    const int kcol = kcol0 + (lane & 15);
+                    const int arow0 = rs * 32 + (lane >> 4) * 8;
+                    // all 8 rows of this fragment share (row>>3), so the
+                    // swizzle key (and the whole LDS column base) hoists
+                    const int bkey = (((arow0 >> 3) & 1) << 1);
+                    const int bg = (kcol >> 3) ^ bkey;
+                    const unsigned short* col =
+                        &x_lds[cur * x_buf + arow0 * x_stride + bg * 8 + (kcol & 7)];
+#pragma unroll
+                    for (int j = 0; j < 8; ++j)
+                        a.u[j] = col[j * x_stride];
+                    b.q = *(U4*)&rt_lds[(lane & 15) * rt_stride + rs * 32 + (lane >> 4) * 8];
+                    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, acc, 0, 0, 0);
+                }
+                g_acc[c * 2 + t2] = acc;
+            }
+            cur ^= 1;
+        }
+        __syncthreads();  // rt_lds reuse next tile
+#undef LOAD_CHUNK2
+#undef WRITE_CHUNK2
+    }
+
+    // ---- epilogue: block partials -> slab (same layout as v1) ----
+    red_lds[threadIdx.x] = logp_acc;
+    __syncthreads();
+    float* slab_blk = slab + (long long)blockIdx.x * (BCH + (long long)K * BCH);
+    if (threadIdx.x < BCH) {
+        float s = 0.f;
+        for (int i = threadIdx.x; i < 256; i += BCH) s += red_lds[i];
+        slab_blk[threadIdx.x] = s;
+    }
+    float* g_slab = slab_blk + BCH;
+#pragma unroll
+    for (int c = 0; c < n_chunks; ++c) {
+#pragma unroll
+        for (int t2 = 0; t2 < 2; ++t2) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int kcol = c * BL_CHUNK + wid * 32 + t2 * 16 + (lane >> 4) * 4 + r;
+                const int chain = lane & 15;
+                g_slab[(long long)kcol * BCH + chain] = g_acc[c * 2 + t2][r];
+            }
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
 // Batched logistic, LDS-resident-tile variant: phase B never re-reads HBM
 // ---------------------------------------------------------------------------
 // The chunked variant's phase-B re-read misses L2 (2 blocks/CU x 32 CUs x
@@ -1304,8 +1497,12 @@ extern "C" int fed_logistic_glm_batched(
     hipStream_t stream = (hipStream_t)stream_v;
     if (K != 512 && K != 1024) return -4;
     const int block = 256;
+    const char* v1_env = getenv("FED_BATCHED_V1");
+    const bool v1_variant = v1_env && atoi(v1_env) != 0;
+    // v2 runs 3 blocks/CU (~38 KB LDS); 768 blocks covers every CU thrice
+    const int grid_cap = v1_variant ? 304 : 768;  // v2: 3 blocks/CU
     int grid = pick_grid(n_rows / BL_ROWS + 1, 1);
-    if (grid > 304) grid = 304;  // slab size cap; >256 CUs covered
+    if (grid > grid_cap) grid = grid_cap;
     const long long slab_cols = BCH + (long long)K * BCH;
     if ((long long)grid * slab_cols * 4 > ws_bytes)
         grid = (int)(ws_bytes / (slab_cols * 4));
@@ -1313,9 +1510,23 @@ extern "C" int fed_logistic_glm_batched(
     const int lds_bytes =
         (BCH * (K + TPAD) + 2 * BL_ROWS * (BL_CHUNK + XPAD) + BCH * (BL_ROWS + RPAD) + 8) * 2 +
         (BL_ROWS + 256) * 4 + 64;
+    const int lds_bytes_v2 =
+        (2 * BL_ROWS * (BL_CHUNK + XPAD) + BCH * (BL_ROWS + RPAD) + 8) * 2 +
+        (BL_ROWS + 256) * 4 + 64;
     const char* lds_env = getenv("FED_BATCHED_LDS");
     const bool lds_variant = lds_env && atoi(lds_env) != 0;
-    if (lds_variant) {
+    if (!lds_variant && !v1_variant) {
+        if (K == 1024)
+            hipLaunchKernelGGL(k_logistic_glm_batched_v2<1024>, dim3(grid), dim3(block),
+                               lds_bytes_v2, stream, (const unsigned short*)X,
+                               (const unsigned short*)y, n_rows,
+                               (const unsigned short*)theta_t_bf16, workspace);
+        else
+            hipLaunchKernelGGL(k_logistic_glm_batched_v2<512>, dim3(grid), dim3(block),
+                               lds_bytes_v2, stream, (const unsigned short*)X,
+                               (const unsigned short*)y, n_rows,
+                               (const unsigned short*)theta_t_bf16, workspace);
+    } else if (lds_variant) {
         // whole-tile-resident variant: 1 block/CU, dynamic LDS ~132 KB
         const int lds2 = (BL_ROWS * (K + XPAD) + BCH * (BL_ROWS + RPAD) + 8) * 2 +
                          (BL_ROWS + 256) * 4 + 64;
