@@ -67,6 +67,18 @@ def build(force: bool = False, verbose: bool = True) -> Path:
             "-o",
             str(WORKER_BIN),
         ]
+        # gRPC edge: HTTP/2 framing + HPACK via libnghttp2 when the image
+        # provides it (header from conda, ABI-stable runtime lib from the
+        # system); without it the worker builds with the fast transport only.
+        nghttp2_inc = Path("/opt/conda/include/nghttp2/nghttp2.h")
+        nghttp2_lib = Path("/usr/lib/x86_64-linux-gnu/libnghttp2.so.14")
+        if nghttp2_inc.exists() and nghttp2_lib.exists():
+            cmd[cmd.index(str(WORKER_SRC)) : cmd.index(str(WORKER_SRC)) + 1] = [
+                "-I/opt/conda/include",
+                str(WORKER_SRC),
+                f"-L{nghttp2_lib.parent}",
+                f"-l:{nghttp2_lib.name}",
+            ]
         if verbose:
             print("[pytensor_federated_amd.ops.build]", " ".join(cmd))
         subprocess.run(cmd, check=True)

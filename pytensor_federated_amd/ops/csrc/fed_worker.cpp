@@ -196,7 +196,8 @@ typedef void* (*host_alloc_fn_t)(long long);
 typedef void* (*dev_alloc_fn_t)(long long);
 
 enum { FED_F32 = 0, FED_F64 = 1, FED_BF16 = 2 };
-enum { MODEL_LINEAR = 0, MODEL_LOGISTIC = 1, MODEL_ODE = 2 };
+enum { MODEL_LINEAR = 0, MODEL_LOGISTIC = 1, MODEL_ODE = 2,
+       MODEL_ECHO = 3 /* dev/CI: sum of scalar inputs, no GPU */ };
 
 struct Worker {
     void* x_dev = nullptr;
@@ -222,7 +223,160 @@ struct Worker {
     double* theta_dev = nullptr;  // f64[4]
     unsigned long long seq = 0;
     std::atomic<int> n_clients{0};  // touched from every client thread
+    std::mutex eval_mu;             // one evaluation in flight at a time
 };
+
+// ---------------------------------------------------------------------------
+// transport-independent request handlers (shared by FEDS1 and gRPC edges)
+// ---------------------------------------------------------------------------
+
+// Evaluate one InputArrays payload against the worker's model.  On success,
+// ``out`` receives the serialized OutputArrays (uuid echoed) and true is
+// returned; on failure ``err`` carries a message and false is returned.
+static bool evaluate_payload(Worker& w, const unsigned char* payload, size_t ln,
+                             std::string& out, std::string& err) {
+    std::vector<ParsedArray> items;
+    std::string uuid;
+    if (w.model == MODEL_ECHO) {
+        // transport self-test: out = [sum of scalar inputs]; lets the HTTP/2
+        // and FEDS1 edges be exercised end-to-end on a GPU-less CI box
+        if (!parse_input_arrays(payload, ln, items, uuid) || items.empty()) {
+            err = "expected at least one scalar input";
+            return false;
+        }
+        double s = 0.0;
+        for (const auto& it : items) s += scalar_value(it);
+        put_len_field(out, 1, encode_f64_scalar(s));
+        put_len_field(out, 2, uuid);
+        return true;
+    }
+    if (w.model == MODEL_LINEAR) {
+        if (!parse_input_arrays(payload, ln, items, uuid) || items.size() != 2) {
+            err = "expected 2 scalar inputs (intercept, slope)";
+            return false;
+        }
+        const double a = scalar_value(items[0]);
+        const double b = scalar_value(items[1]);
+        int rc;
+        double res[3];
+        {
+            std::lock_guard<std::mutex> lock(w.eval_mu);
+            w.seq++;
+            rc = w.eval(w.x_dev, w.y_dev, w.n, a, b, w.sigma, w.out_dev, w.mailbox,
+                        w.ws_dev, (72 + 3 * 2048) * 8, w.dtype, nullptr, w.seq);
+            res[0] = w.mailbox[0]; res[1] = w.mailbox[1]; res[2] = w.mailbox[2];
+        }
+        if (rc != 0) {
+            char msg[64];
+            snprintf(msg, sizeof(msg), "kernel eval failed (%d)", rc);
+            err = msg;
+            return false;
+        }
+        put_len_field(out, 1, encode_f64_scalar(res[0]));  // logp
+        put_len_field(out, 1, encode_f64_scalar(res[1]));  // d/da
+        put_len_field(out, 1, encode_f64_scalar(res[2]));  // d/db
+        put_len_field(out, 2, uuid);
+        return true;
+    }
+    if (w.model == MODEL_ODE) {
+        if (!parse_input_arrays(payload, ln, items, uuid) || items.size() != 1 ||
+            items[0].dtype != "float64" || items[0].data.size() != 4 * 8) {
+            err = "expected one float64 theta[4] input";
+            return false;
+        }
+        {
+            std::lock_guard<std::mutex> lock(w.eval_mu);
+            int rc = (int)hipMemcpy(w.theta_dev, items[0].data.data(), 4 * 8,
+                                    hipMemcpyHostToDevice);
+            if (rc == 0)
+                rc = w.eval_ode((const double*)w.x_dev, (const double*)w.y_dev,
+                                w.obs_dev, w.ode_steps, w.ode_B, w.ode_h,
+                                w.sigma, w.theta_dev, w.ws_dev, w.out_dev, nullptr);
+            if (rc == 0) rc = (int)hipDeviceSynchronize();
+            if (rc == 0)
+                rc = (int)hipMemcpy(w.out_host.data(), w.out_dev, 5 * 8,
+                                    hipMemcpyDeviceToHost);
+            if (rc != 0) {
+                err = "ode eval failed";
+                return false;
+            }
+            put_len_field(out, 1, encode_f64_scalar(w.out_host[0] + w.ode_logp_const));
+            put_len_field(out, 1, encode_f64_vector(&w.out_host[1], 4));
+        }
+        put_len_field(out, 2, uuid);
+        return true;
+    }
+    // logistic GLM: beta[K] f64 in
+    if (!parse_input_arrays(payload, ln, items, uuid) || items.size() != 1 ||
+        items[0].dtype != "float64" || items[0].data.size() != (size_t)w.K * 8) {
+        err = "expected one float64 beta[K] input";
+        return false;
+    }
+    {
+        std::lock_guard<std::mutex> lock(w.eval_mu);
+        std::vector<float> beta32(w.K);
+        const double* bd = (const double*)items[0].data.data();
+        for (int i = 0; i < w.K; ++i) beta32[i] = (float)bd[i];
+        if (hipMemcpy(w.beta_dev, beta32.data(), w.K * 4, hipMemcpyHostToDevice) !=
+            hipSuccess) {
+            err = "beta upload failed";
+            return false;
+        }
+        int rc = w.eval_logistic(w.x_dev, w.y_dev, w.n, w.K, w.beta_dev, w.out_dev,
+                                 w.ws_f32, (long long)1024 * w.K * 4, w.dtype, nullptr);
+        if (rc == 0) rc = (int)hipDeviceSynchronize();
+        if (rc != 0) {
+            err = "logistic eval failed";
+            return false;
+        }
+        if (hipMemcpy(w.out_host.data(), w.out_dev, (1 + w.K) * 8,
+                      hipMemcpyDeviceToHost) != hipSuccess) {
+            err = "result readback failed";
+            return false;
+        }
+        put_len_field(out, 1, encode_f64_scalar(w.out_host[0]));        // logp
+        put_len_field(out, 1, encode_f64_vector(&w.out_host[1], w.K));  // grad
+    }
+    put_len_field(out, 2, uuid);
+    return true;
+}
+
+// Serialize a GetLoadResult with GPU-first telemetry (mirrors the Python
+// edge's service.determine_load, service.py:141-155: percent_cpu carries
+// GPU busy %, percent_ram carries VRAM use %).
+static std::string get_load_payload(Worker& w) {
+    std::string out;
+    put_varint(out, (1 << 3) | 0);  // n_clients
+    put_varint(out, (unsigned long long)w.n_clients.load());
+    float busy = -1.0f;
+    for (int card = 0; card < 8 && busy < 0; ++card) {
+        char path[64];
+        snprintf(path, sizeof(path),
+                 "/sys/class/drm/card%d/device/gpu_busy_percent", card);
+        FILE* f = fopen(path, "r");
+        if (f) {
+            int v;
+            if (fscanf(f, "%d", &v) == 1) busy = (float)v;
+            fclose(f);
+        }
+    }
+    if (busy < 0) {  // no amdgpu sysfs: host loadavg fallback
+        double la = 0;
+        if (getloadavg(&la, 1) == 1) {
+            long ncpu = sysconf(_SC_NPROCESSORS_ONLN);
+            busy = (float)(100.0 * la / (ncpu > 0 ? ncpu : 1));
+        } else {
+            busy = 0.0f;
+        }
+    }
+    float vram = 0.0f;
+    size_t free_b = 0, total_b = 0;
+    if (hipMemGetInfo(&free_b, &total_b) == hipSuccess && total_b)
+        vram = (float)(100.0 * (1.0 - (double)free_b / (double)total_b));
+    put_float_field(out, 2, busy);
+    put_float_field(out, 3, vram);
+    return out;
+}
 
 static unsigned short f32_to_bf16(float f) {
     unsigned int u;
@@ -428,141 +582,15 @@ static void serve_client(Worker& w, int fd) {
         }
         payload.resize(ln);
         if (ln && !read_exact(fd, payload.data(), ln)) break;
-        static std::mutex eval_mu;
-        if (hdr[0] == 0x01 && w.model == MODEL_LINEAR) {  // Evaluate: gaussian linear
-            std::vector<ParsedArray> items;
-            std::string uuid;
-            if (!parse_input_arrays(payload.data(), ln, items, uuid) || items.size() != 2) {
-                write_frame(fd, 0xFF, "expected 2 scalar inputs (intercept, slope)");
+        if (hdr[0] == 0x01) {  // Evaluate
+            std::string out, err;
+            if (!evaluate_payload(w, payload.data(), ln, out, err)) {
+                write_frame(fd, 0xFF, err);
                 continue;
             }
-            const double a = scalar_value(items[0]);
-            const double b = scalar_value(items[1]);
-            int rc;
-            double res[3];
-            {
-                std::lock_guard<std::mutex> lock(eval_mu);
-                w.seq++;
-                rc = w.eval(w.x_dev, w.y_dev, w.n, a, b, w.sigma, w.out_dev, w.mailbox,
-                            w.ws_dev, (72 + 3 * 2048) * 8, w.dtype, nullptr, w.seq);
-                res[0] = w.mailbox[0]; res[1] = w.mailbox[1]; res[2] = w.mailbox[2];
-            }
-            if (rc != 0) {
-                char msg[64];
-                snprintf(msg, sizeof(msg), "kernel eval failed (%d)", rc);
-                write_frame(fd, 0xFF, msg);
-                continue;
-            }
-            std::string out;
-            put_len_field(out, 1, encode_f64_scalar(res[0]));  // logp
-            put_len_field(out, 1, encode_f64_scalar(res[1]));  // d/da
-            put_len_field(out, 1, encode_f64_scalar(res[2]));  // d/db
-            put_len_field(out, 2, uuid);
-            if (!write_frame(fd, 0x81, out)) break;
-        } else if (hdr[0] == 0x01 && w.model == MODEL_ODE) {  // theta[4] f64 in
-            std::vector<ParsedArray> items;
-            std::string uuid;
-            if (!parse_input_arrays(payload.data(), ln, items, uuid) || items.size() != 1 ||
-                items[0].dtype != "float64" || items[0].data.size() != 4 * 8) {
-                write_frame(fd, 0xFF, "expected one float64 theta[4] input");
-                continue;
-            }
-            std::string out;
-            {
-                std::lock_guard<std::mutex> lock(eval_mu);
-                int rc = (int)hipMemcpy(w.theta_dev, items[0].data.data(), 4 * 8,
-                                        hipMemcpyHostToDevice);
-                if (rc == 0)
-                    rc = w.eval_ode((const double*)w.x_dev, (const double*)w.y_dev,
-                                    w.obs_dev, w.ode_steps, w.ode_B, w.ode_h,
-                                    w.sigma, w.theta_dev, w.ws_dev, w.out_dev,
-                                    nullptr);
-                if (rc == 0) rc = (int)hipDeviceSynchronize();
-                if (rc == 0)
-                    rc = (int)hipMemcpy(w.out_host.data(), w.out_dev, 5 * 8,
-                                        hipMemcpyDeviceToHost);
-                if (rc != 0) {
-                    write_frame(fd, 0xFF, "ode eval failed");
-                    continue;
-                }
-                put_len_field(out, 1, encode_f64_scalar(w.out_host[0] + w.ode_logp_const));
-                put_len_field(out, 1, encode_f64_vector(&w.out_host[1], 4));
-            }
-            put_len_field(out, 2, uuid);
-            if (!write_frame(fd, 0x81, out)) break;
-        } else if (hdr[0] == 0x01) {  // Evaluate: logistic GLM, beta[K] f64 in
-            std::vector<ParsedArray> items;
-            std::string uuid;
-            if (!parse_input_arrays(payload.data(), ln, items, uuid) || items.size() != 1 ||
-                items[0].dtype != "float64" ||
-                items[0].data.size() != (size_t)w.K * 8) {
-                write_frame(fd, 0xFF, "expected one float64 beta[K] input");
-                continue;
-            }
-            std::string out;
-            {
-                std::lock_guard<std::mutex> lock(eval_mu);
-                std::vector<float> beta32(w.K);
-                const double* bd = (const double*)items[0].data.data();
-                for (int i = 0; i < w.K; ++i) beta32[i] = (float)bd[i];
-                if (hipMemcpy(w.beta_dev, beta32.data(), w.K * 4,
-                              hipMemcpyHostToDevice) != hipSuccess) {
-                    write_frame(fd, 0xFF, "beta upload failed");
-                    continue;
-                }
-                int rc = w.eval_logistic(w.x_dev, w.y_dev, w.n, w.K, w.beta_dev,
-                                         w.out_dev, w.ws_f32,
-                                         (long long)1024 * w.K * 4, w.dtype, nullptr);
-                if (rc == 0) rc = (int)hipDeviceSynchronize();
-                if (rc != 0) {
-                    write_frame(fd, 0xFF, "logistic eval failed");
-                    continue;
-                }
-                if (hipMemcpy(w.out_host.data(), w.out_dev, (1 + w.K) * 8,
-                              hipMemcpyDeviceToHost) != hipSuccess) {
-                    write_frame(fd, 0xFF, "result readback failed");
-                    continue;
-                }
-                put_len_field(out, 1, encode_f64_scalar(w.out_host[0]));          // logp
-                put_len_field(out, 1, encode_f64_vector(&w.out_host[1], w.K));    // grad
-            }
-            put_len_field(out, 2, uuid);
             if (!write_frame(fd, 0x81, out)) break;
         } else if (hdr[0] == 0x02) {  // GetLoad
-            // GPU-first telemetry, mirroring the Python edge's
-            // service.determine_load (service.py:141-155): percent_cpu
-            // carries GPU busy %, percent_ram carries VRAM use %.
-            std::string out;
-            put_varint(out, (1 << 3) | 0);  // n_clients
-            put_varint(out, (unsigned long long)w.n_clients.load());
-            float busy = -1.0f;
-            for (int card = 0; card < 8 && busy < 0; ++card) {
-                char path[64];
-                snprintf(path, sizeof(path),
-                         "/sys/class/drm/card%d/device/gpu_busy_percent", card);
-                FILE* f = fopen(path, "r");
-                if (f) {
-                    int v;
-                    if (fscanf(f, "%d", &v) == 1) busy = (float)v;
-                    fclose(f);
-                }
-            }
-            if (busy < 0) {  // no amdgpu sysfs: host loadavg fallback
-                double la = 0;
-                if (getloadavg(&la, 1) == 1) {
-                    long ncpu = sysconf(_SC_NPROCESSORS_ONLN);
-                    busy = (float)(100.0 * la / (ncpu > 0 ? ncpu : 1));
-                } else {
-                    busy = 0.0f;
-                }
-            }
-            float vram = 0.0f;
-            size_t free_b = 0, total_b = 0;
-            if (hipMemGetInfo(&free_b, &total_b) == hipSuccess && total_b)
-                vram = (float)(100.0 * (1.0 - (double)free_b / (double)total_b));
-            put_float_field(out, 2, busy);
-            put_float_field(out, 3, vram);
-            if (!write_frame(fd, 0x82, out)) break;
+            if (!write_frame(fd, 0x82, get_load_payload(w))) break;
         } else {
             write_frame(fd, 0xFF, "unknown frame type");
         }
@@ -572,18 +600,302 @@ static void serve_client(Worker& w, int fd) {
     close(fd);
 }
 
+// ---------------------------------------------------------------------------
+// gRPC edge (HTTP/2 via libnghttp2): the reference's language-portable
+// protocol face (reference README.md:35, service.py:75-115).  The three
+// routes of service.proto are served natively -- a betterproto/grpclib or
+// grpcio client evaluates against this worker with no Python sidecar.
+// ---------------------------------------------------------------------------
+#if __has_include(<nghttp2/nghttp2.h>)
+#define FED_HAVE_NGHTTP2 1
+#include <nghttp2/nghttp2.h>
+
+namespace fedgrpc {
+
+enum Route { R_UNKNOWN = 0, R_EVALUATE, R_EVALUATE_STREAM, R_GET_LOAD };
+
+struct StreamCtx {
+    Route route = R_UNKNOWN;
+    std::string inbuf;    // accumulated DATA bytes (gRPC length-prefixed)
+    std::string outbuf;   // framed response messages not yet sent
+    size_t out_off = 0;
+    bool headers_sent = false;
+    bool client_done = false;   // client half-closed (END_STREAM seen)
+    bool failed = false;
+    int grpc_status = 0;        // trailer status
+    std::string grpc_message;
+};
+
+struct ConnCtx {
+    Worker* w;
+    int fd;
+    nghttp2_session* session = nullptr;
+};
+
+static ssize_t data_read_cb(nghttp2_session* session, int32_t stream_id,
+                            uint8_t* buf, size_t length, uint32_t* data_flags,
+                            nghttp2_data_source* source, void* /*user*/) {
+    StreamCtx* sc = (StreamCtx*)source->ptr;
+    const size_t avail = sc->outbuf.size() - sc->out_off;
+    if (avail == 0) {
+        if (sc->client_done || sc->failed) {
+            // all responses sent: close the data sequence, then trailers
+            *data_flags |= NGHTTP2_DATA_FLAG_EOF | NGHTTP2_DATA_FLAG_NO_END_STREAM;
+            char status[8];
+            snprintf(status, sizeof(status), "%d", sc->grpc_status);
+            std::vector<nghttp2_nv> trailers;
+            nghttp2_nv st = {(uint8_t*)"grpc-status", (uint8_t*)status,
+                             11, strlen(status), NGHTTP2_NV_FLAG_NONE};
+            trailers.push_back(st);
+            nghttp2_nv msg = {(uint8_t*)"grpc-message",
+                              (uint8_t*)sc->grpc_message.c_str(), 12,
+                              sc->grpc_message.size(), NGHTTP2_NV_FLAG_NONE};
+            if (!sc->grpc_message.empty()) trailers.push_back(msg);
+            nghttp2_submit_trailer(session, stream_id, trailers.data(),
+                                   trailers.size());
+            return 0;
+        }
+        return NGHTTP2_ERR_DEFERRED;  // resumed when the next reply is queued
+    }
+    const size_t n = avail < length ? avail : length;
+    memcpy(buf, sc->outbuf.data() + sc->out_off, n);
+    sc->out_off += n;
+    if (sc->out_off == sc->outbuf.size()) {
+        sc->outbuf.clear();
+        sc->out_off = 0;
+    }
+    return (ssize_t)n;
+}
+
+static void queue_grpc_message(StreamCtx* sc, const std::string& payload) {
+    char hdr[5];
+    hdr[0] = 0;  // uncompressed
+    const uint32_t ln = (uint32_t)payload.size();
+    hdr[1] = (char)(ln >> 24); hdr[2] = (char)(ln >> 16);
+    hdr[3] = (char)(ln >> 8);  hdr[4] = (char)ln;
+    sc->outbuf.append(hdr, 5);
+    sc->outbuf += payload;
+}
+
+static void ensure_response_started(ConnCtx* cc, int32_t stream_id, StreamCtx* sc) {
+    if (sc->headers_sent) {
+        nghttp2_session_resume_data(cc->session, stream_id);
+        return;
+    }
+    sc->headers_sent = true;
+    static const nghttp2_nv hdrs[] = {
+        {(uint8_t*)":status", (uint8_t*)"200", 7, 3, NGHTTP2_NV_FLAG_NONE},
+        {(uint8_t*)"content-type", (uint8_t*)"application/grpc", 12, 16,
+         NGHTTP2_NV_FLAG_NONE},
+    };
+    nghttp2_data_provider prov;
+    prov.source.ptr = sc;
+    prov.read_callback = data_read_cb;
+    nghttp2_submit_response(cc->session, stream_id, hdrs, 2, &prov);
+}
+
+static void fail_stream(ConnCtx* cc, int32_t stream_id, StreamCtx* sc,
+                        int status, const std::string& message) {
+    sc->failed = true;
+    sc->grpc_status = status;
+    sc->grpc_message = message;
+    ensure_response_started(cc, stream_id, sc);
+}
+
+// process complete length-prefixed gRPC messages accumulated in inbuf
+static void process_messages(ConnCtx* cc, int32_t stream_id, StreamCtx* sc) {
+    while (!sc->failed && sc->inbuf.size() >= 5) {
+        const unsigned char* p = (const unsigned char*)sc->inbuf.data();
+        if (p[0] != 0) {  // compressed messages unsupported
+            fail_stream(cc, stream_id, sc, 12, "message compression unsupported");
+            return;
+        }
+        const uint32_t ln = ((uint32_t)p[1] << 24) | ((uint32_t)p[2] << 16) |
+                            ((uint32_t)p[3] << 8) | (uint32_t)p[4];
+        if (ln > 256u * 1024 * 1024) {
+            fail_stream(cc, stream_id, sc, 8, "message exceeds size cap");
+            return;
+        }
+        if (sc->inbuf.size() < 5ull + ln) return;  // incomplete
+        std::string out, err;
+        bool ok;
+        if (sc->route == R_GET_LOAD) {
+            out = get_load_payload(*cc->w);
+            ok = true;
+        } else {
+            ok = evaluate_payload(*cc->w, p + 5, ln, out, err);
+        }
+        sc->inbuf.erase(0, 5ull + ln);
+        if (!ok) {
+            fail_stream(cc, stream_id, sc, 13, err);
+            return;
+        }
+        queue_grpc_message(sc, out);
+        if (sc->route != R_EVALUATE_STREAM)
+            sc->client_done = true;  // unary: one message, then trailers
+        ensure_response_started(cc, stream_id, sc);
+    }
+}
+
+static int on_begin_headers_cb(nghttp2_session* session,
+                               const nghttp2_frame* frame, void* /*user*/) {
+    if (frame->hd.type != NGHTTP2_HEADERS ||
+        frame->headers.cat != NGHTTP2_HCAT_REQUEST)
+        return 0;
+    nghttp2_session_set_stream_user_data(session, frame->hd.stream_id,
+                                         new StreamCtx());
+    return 0;
+}
+
+static int on_header_cb(nghttp2_session* session, const nghttp2_frame* frame,
+                        const uint8_t* name, size_t namelen,
+                        const uint8_t* value, size_t valuelen, uint8_t /*flags*/,
+                        void* /*user*/) {
+    if (frame->hd.type != NGHTTP2_HEADERS) return 0;
+    StreamCtx* sc =
+        (StreamCtx*)nghttp2_session_get_stream_user_data(session, frame->hd.stream_id);
+    if (!sc) return 0;
+    if (namelen == 5 && memcmp(name, ":path", 5) == 0) {
+        const std::string path((const char*)value, valuelen);
+        if (path == "/ArraysToArraysService/Evaluate") sc->route = R_EVALUATE;
+        else if (path == "/ArraysToArraysService/EvaluateStream")
+            sc->route = R_EVALUATE_STREAM;
+        else if (path == "/ArraysToArraysService/GetLoad") sc->route = R_GET_LOAD;
+    }
+    return 0;
+}
+
+static int on_data_chunk_cb(nghttp2_session* /*session*/, uint8_t /*flags*/,
+                            int32_t stream_id, const uint8_t* data, size_t len,
+                            void* user) {
+    ConnCtx* cc = (ConnCtx*)user;
+    StreamCtx* sc =
+        (StreamCtx*)nghttp2_session_get_stream_user_data(cc->session, stream_id);
+    if (!sc) return 0;
+    sc->inbuf.append((const char*)data, len);
+    process_messages(cc, stream_id, sc);
+    return 0;
+}
+
+static int on_frame_recv_cb(nghttp2_session* session, const nghttp2_frame* frame,
+                            void* user) {
+    ConnCtx* cc = (ConnCtx*)user;
+    StreamCtx* sc =
+        (StreamCtx*)nghttp2_session_get_stream_user_data(session, frame->hd.stream_id);
+    if (!sc) return 0;
+    if (frame->hd.type == NGHTTP2_HEADERS &&
+        frame->headers.cat == NGHTTP2_HCAT_REQUEST && sc->route == R_UNKNOWN) {
+        fail_stream(cc, frame->hd.stream_id, sc, 12, "unknown method");
+        return 0;
+    }
+    if ((frame->hd.type == NGHTTP2_DATA || frame->hd.type == NGHTTP2_HEADERS) &&
+        (frame->hd.flags & NGHTTP2_FLAG_END_STREAM)) {
+        sc->client_done = true;  // half-closed: flush replies, then trailers
+        ensure_response_started(cc, frame->hd.stream_id, sc);
+    }
+    return 0;
+}
+
+static int on_stream_close_cb(nghttp2_session* session, int32_t stream_id,
+                              uint32_t /*error_code*/, void* /*user*/) {
+    StreamCtx* sc = (StreamCtx*)nghttp2_session_get_stream_user_data(session, stream_id);
+    if (sc) {
+        delete sc;
+        nghttp2_session_set_stream_user_data(session, stream_id, nullptr);
+    }
+    return 0;
+}
+
+static bool write_all(int fd, const uint8_t* p, size_t n) {
+    while (n) {
+        ssize_t w = write(fd, p, n);
+        if (w <= 0) return false;
+        p += w;
+        n -= (size_t)w;
+    }
+    return true;
+}
+
+static void serve_grpc_client(Worker& w, int fd) {
+    int one = 1;
+    setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+    ConnCtx cc;
+    cc.w = &w;
+    cc.fd = fd;
+    nghttp2_session_callbacks* cbs;
+    nghttp2_session_callbacks_new(&cbs);
+    nghttp2_session_callbacks_set_on_begin_headers_callback(cbs, on_begin_headers_cb);
+    nghttp2_session_callbacks_set_on_header_callback(cbs, on_header_cb);
+    nghttp2_session_callbacks_set_on_data_chunk_recv_callback(cbs, on_data_chunk_cb);
+    nghttp2_session_callbacks_set_on_frame_recv_callback(cbs, on_frame_recv_cb);
+    nghttp2_session_callbacks_set_on_stream_close_callback(cbs, on_stream_close_cb);
+    nghttp2_session_server_new(&cc.session, cbs, &cc);
+    nghttp2_session_callbacks_del(cbs);
+
+    nghttp2_settings_entry settings[] = {
+        {NGHTTP2_SETTINGS_MAX_CONCURRENT_STREAMS, 128},
+        {NGHTTP2_SETTINGS_INITIAL_WINDOW_SIZE, 1 << 20},
+    };
+    nghttp2_submit_settings(cc.session, NGHTTP2_FLAG_NONE, settings, 2);
+    nghttp2_session_set_local_window_size(cc.session, NGHTTP2_FLAG_NONE, 0, 1 << 24);
+
+    ++w.n_clients;
+    fprintf(stderr, "grpc client connected (now %d)\n", w.n_clients.load());
+    uint8_t buf[65536];
+    while (true) {
+        while (nghttp2_session_want_write(cc.session)) {
+            const uint8_t* out = nullptr;
+            ssize_t n = nghttp2_session_mem_send(cc.session, &out);
+            if (n <= 0) break;
+            if (!write_all(fd, out, (size_t)n)) goto done;
+        }
+        if (!nghttp2_session_want_read(cc.session) &&
+            !nghttp2_session_want_write(cc.session))
+            break;
+        ssize_t r = read(fd, buf, sizeof(buf));
+        if (r <= 0) break;
+        if (nghttp2_session_mem_recv(cc.session, buf, (size_t)r) < 0) break;
+    }
+done:
+    fprintf(stderr, "grpc client disconnected (now %d)\n", w.n_clients.load() - 1);
+    --w.n_clients;
+    nghttp2_session_del(cc.session);
+    close(fd);
+}
+
+}  // namespace fedgrpc
+#endif  // nghttp2
+
+static int listen_on(int port) {
+    int srv = socket(AF_INET, SOCK_STREAM, 0);
+    int one = 1;
+    setsockopt(srv, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+    sockaddr_in addr{};
+    addr.sin_family = AF_INET;
+    addr.sin_addr.s_addr = htonl(INADDR_LOOPBACK);
+    addr.sin_port = htons((unsigned short)port);
+    if (bind(srv, (sockaddr*)&addr, sizeof(addr)) != 0 || listen(srv, 8) != 0) {
+        fprintf(stderr, "bind/listen on %d failed\n", port);
+        return -1;
+    }
+    return srv;
+}
+
 int main(int argc, char** argv) {
     int port = 9600;
+    int grpc_port = 0;
     const char* data_path = nullptr;
     Worker w;
     for (int i = 1; i < argc - 1; ++i) {
         if (!strcmp(argv[i], "--port")) port = atoi(argv[++i]);
+        else if (!strcmp(argv[i], "--grpc-port")) grpc_port = atoi(argv[++i]);
         else if (!strcmp(argv[i], "--data")) data_path = argv[++i];
         else if (!strcmp(argv[i], "--sigma")) w.sigma = atof(argv[++i]);
         else if (!strcmp(argv[i], "--model")) {
             const char* m = argv[++i];
             w.model = !strcmp(m, "logistic") ? MODEL_LOGISTIC
                     : !strcmp(m, "ode")      ? MODEL_ODE
+                    : !strcmp(m, "echo")     ? MODEL_ECHO
                                              : MODEL_LINEAR;
         }
         else if (!strcmp(argv[i], "--dtype")) {
@@ -591,12 +903,42 @@ int main(int argc, char** argv) {
             w.dtype = !strcmp(d, "f64") ? FED_F64 : !strcmp(d, "f32") ? FED_F32 : FED_BF16;
         }
     }
-    if (!data_path) {
+    if (!data_path && w.model != MODEL_ECHO) {
         fprintf(stderr, "usage: fed_worker --port P --data shard.bin "
-                        "[--model linear|logistic|ode] [--sigma S] [--dtype bf16|f32|f64]\n");
+                        "[--model linear|logistic|ode|echo] [--grpc-port G] "
+                        "[--sigma S] [--dtype bf16|f32|f64]\n");
         return 2;
     }
     signal(SIGPIPE, SIG_IGN);
+
+    if (w.model == MODEL_ECHO) {
+        int srv = listen_on(port);
+        if (srv < 0) return 2;
+        fprintf(stderr, "fed_worker echo mode on 127.0.0.1:%d\n", port);
+        if (grpc_port > 0) {
+#ifdef FED_HAVE_NGHTTP2
+            int gsrv = listen_on(grpc_port);
+            if (gsrv < 0) return 2;
+            fprintf(stderr, "fed_worker echo gRPC on 127.0.0.1:%d\n", grpc_port);
+            std::thread([&w, gsrv] {
+                while (true) {
+                    int fd = accept(gsrv, nullptr, nullptr);
+                    if (fd < 0) continue;
+                    std::thread([&w, fd] { fedgrpc::serve_grpc_client(w, fd); }).detach();
+                }
+            }).detach();
+#else
+            fprintf(stderr, "--grpc-port requires libnghttp2 at build time\n");
+            return 2;
+#endif
+        }
+        while (true) {
+            int fd = accept(srv, nullptr, nullptr);
+            if (fd < 0) continue;
+            std::thread([&w, fd] { serve_client(w, fd); }).detach();
+        }
+        return 0;
+    }
 
     // kernels
     const char* lib_env = getenv("FEDOPS_LIB");
@@ -642,18 +984,26 @@ int main(int argc, char** argv) {
     }
     w.mailbox[3] = 0.0;
 
-    int srv = socket(AF_INET, SOCK_STREAM, 0);
-    int one = 1;
-    setsockopt(srv, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
-    sockaddr_in addr{};
-    addr.sin_family = AF_INET;
-    addr.sin_addr.s_addr = htonl(INADDR_LOOPBACK);
-    addr.sin_port = htons((unsigned short)port);
-    if (bind(srv, (sockaddr*)&addr, sizeof(addr)) != 0 || listen(srv, 8) != 0) {
-        fprintf(stderr, "bind/listen on %d failed\n", port);
+    int srv = listen_on(port);
+    if (srv < 0) return 2;
+    fprintf(stderr, "fed_worker serving %lld rows on 127.0.0.1:%d (fast)\n", w.n, port);
+    if (grpc_port > 0) {
+#ifdef FED_HAVE_NGHTTP2
+        int gsrv = listen_on(grpc_port);
+        if (gsrv < 0) return 2;
+        fprintf(stderr, "fed_worker serving gRPC on 127.0.0.1:%d\n", grpc_port);
+        std::thread([&w, gsrv] {
+            while (true) {
+                int fd = accept(gsrv, nullptr, nullptr);
+                if (fd < 0) continue;
+                std::thread([&w, fd] { fedgrpc::serve_grpc_client(w, fd); }).detach();
+            }
+        }).detach();
+#else
+        fprintf(stderr, "--grpc-port requires libnghttp2 at build time\n");
         return 2;
+#endif
     }
-    fprintf(stderr, "fed_worker serving %lld rows on 127.0.0.1:%d\n", w.n, port);
     while (true) {
         int fd = accept(srv, nullptr, nullptr);
         if (fd < 0) continue;

@@ -1166,9 +1166,6 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched(
 //    group-bit-1 flipped against each other, separating their banks for
 //    the phase-B scalar reads while leaving the b128 write/read patterns
 //    near-conflict-free.
-//  * Theta fragments read straight from global (32 KB, L2-resident),
-//    freeing 33 KB LDS: the block's footprint drops to ~38 KB, so THREE
-//    blocks fit per CU (12 waves in flight vs v1's 8).
 //  * out-of-range rows handled by address clamping (their logp/resid
 //    contributions are masked later) instead of per-element zero-fill
 //    branches, which de-pipeline hipcc's load scheduling.
@@ -1186,13 +1183,25 @@ __global__ __launch_bounds__(256, 3) void k_logistic_glm_batched_v2(
     constexpr int n_chunks = K / BL_CHUNK;
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    unsigned short* x_lds = (unsigned short*)smem;              // [2][BL_ROWS][BL_CHUNK+XPAD]
+    unsigned short* th_lds = (unsigned short*)smem;             // [BCH][K+TPAD]
+    const int th_stride = K + TPAD;
+    unsigned short* x_lds = th_lds + BCH * th_stride;           // [2][BL_ROWS][BL_CHUNK+XPAD]
     const int x_stride = BL_CHUNK + XPAD;
     const int x_buf = BL_ROWS * x_stride;
     unsigned short* rt_lds = x_lds + 2 * x_buf;                 // [BCH][BL_ROWS+RPAD]
     const int rt_stride = BL_ROWS + RPAD;
     float* y_lds = (float*)(rt_lds + BCH * rt_stride + 8);      // [BL_ROWS]
     float* red_lds = y_lds + BL_ROWS;                           // [256]
+
+    // Theta^T staged once per block (a theta read from global inside the
+    // MFMA loop poisons the vmcnt pipeline: its wait is FIFO-ordered after
+    // the chunk-prefetch loads, so every chunk step drained its prefetch --
+    // measured 2.97 ms vs 1.56 ms v1 before this was reverted to LDS)
+    for (int idx = threadIdx.x * 8; idx < BCH * K; idx += 256 * 8) {
+        const int b = idx / K;
+        const int k = idx % K;
+        *(U4*)&th_lds[b * th_stride + k] = *(const U4*)&theta_t[b * K + k];
+    }
 
     typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
     typedef __attribute__((ext_vector_type(4))) float f32x4_t;
@@ -1248,7 +1257,7 @@ __global__ __launch_bounds__(256, 3) void k_logistic_glm_batched_v2(
                 const int ag = (ks * 4 + (lane >> 4)) ^ akey;
                 a.q = *(U4*)&x_lds[cur * x_buf + arow * x_stride + ag * 8];
                 const int bk = c * BL_CHUNK + ks * 32 + (lane >> 4) * 8;
-                b.q = *(const U4*)&theta_t[(lane & 15) * K + bk];
+                b.q = *(U4*)&th_lds[(lane & 15) * th_stride + bk];
                 z_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, z_acc, 0, 0, 0);
             }
             cur ^= 1;
@@ -1500,7 +1509,7 @@ extern "C" int fed_logistic_glm_batched(
     const char* v1_env = getenv("FED_BATCHED_V1");
     const bool v1_variant = v1_env && atoi(v1_env) != 0;
     // v2 runs 3 blocks/CU (~38 KB LDS); 768 blocks covers every CU thrice
-    const int grid_cap = v1_variant ? 304 : 768;  // v2: 3 blocks/CU
+    const int grid_cap = v1_variant ? 304 : 512;  // v2: 2 blocks/CU
     int grid = pick_grid(n_rows / BL_ROWS + 1, 1);
     if (grid > grid_cap) grid = grid_cap;
     const long long slab_cols = BCH + (long long)K * BCH;
@@ -1511,7 +1520,7 @@ extern "C" int fed_logistic_glm_batched(
         (BCH * (K + TPAD) + 2 * BL_ROWS * (BL_CHUNK + XPAD) + BCH * (BL_ROWS + RPAD) + 8) * 2 +
         (BL_ROWS + 256) * 4 + 64;
     const int lds_bytes_v2 =
-        (2 * BL_ROWS * (BL_CHUNK + XPAD) + BCH * (BL_ROWS + RPAD) + 8) * 2 +
+        (BCH * (K + TPAD) + 2 * BL_ROWS * (BL_CHUNK + XPAD) + BCH * (BL_ROWS + RPAD) + 8) * 2 +
         (BL_ROWS + 256) * 4 + 64;
     const char* lds_env = getenv("FED_BATCHED_LDS");
     const bool lds_variant = lds_env && atoi(lds_env) != 0;
