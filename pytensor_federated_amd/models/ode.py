@@ -26,7 +26,7 @@ import torch
 
 from .base import LogpGradModel
 
-__all__ = ["ODEModel", "lotka_volterra_rhs", "generate_ode_dataset"]
+__all__ = ["ODEModel", "PolynomialRHS", "lotka_volterra_rhs", "generate_ode_dataset"]
 
 
 def lotka_volterra_rhs(t, u: torch.Tensor, theta: torch.Tensor) -> torch.Tensor:
@@ -36,6 +36,89 @@ def lotka_volterra_rhs(t, u: torch.Tensor, theta: torch.Tensor) -> torch.Tensor:
     dprey = alpha * prey - beta * prey * pred
     dpred = delta * prey * pred - gamma * pred
     return torch.stack([dprey, dpred], dim=-1)
+
+
+class PolynomialRHS:
+    """Coefficient-table polynomial vector field:
+
+        du_d/dt = sum_t  c_t * theta_{j_t} * prod_i u_i^{e_ti}
+
+    (``j_t = -1`` drops the theta factor.)  One native CDNA4 kernel
+    (ops/csrc/ode_poly.hip) interprets the table, so any model of this
+    family -- Lotka-Volterra, SIR, oscillators, mass-action kinetics --
+    gets the in-register forward+adjoint path WITHOUT recompilation;
+    ``__call__`` evaluates the same field in torch for the eager/autograd
+    path and the golden tests.  Limits for the native path: D<=4 states,
+    P<=8 thetas, <=16 terms, integer exponents 0..15.
+    """
+
+    def __init__(self, terms, D: int, P: int):
+        self.terms = [
+            (int(d), int(j), float(c), tuple(int(v) for v in e))
+            for (d, j, c, e) in terms
+        ]
+        self.D = int(D)
+        self.P = int(P)
+        for d, j, c, e in self.terms:
+            if not (0 <= d < self.D) or not (-1 <= j < self.P):
+                raise ValueError(f"bad term (d={d}, j={j}) for D={D}, P={P}")
+            if len(e) > self.D:
+                raise ValueError(f"exponent tuple {e} longer than D={D}")
+
+    def __call__(self, t, u: torch.Tensor, theta: torch.Tensor) -> torch.Tensor:
+        comps = []
+        for d in range(self.D):
+            acc = None
+            for dt, j, c, e in self.terms:
+                if dt != d:
+                    continue
+                mon = None
+                for i, ei in enumerate(e):
+                    if ei:
+                        fac = u[..., i] ** ei if ei > 1 else u[..., i]
+                        mon = fac if mon is None else mon * fac
+                val = mon * c if mon is not None else u.new_full(u.shape[:-1], c)
+                if j >= 0:
+                    val = val * theta[j]
+                acc = val if acc is None else acc + val
+            comps.append(acc if acc is not None else u.new_zeros(u.shape[:-1]))
+        return torch.stack(comps, dim=-1)
+
+    def native_ok(self) -> bool:
+        return (
+            self.D <= 4
+            and self.P <= 8
+            and len(self.terms) <= 16
+            and all(all(0 <= v <= 15 for v in e) for *_x, e in self.terms)
+        )
+
+    @classmethod
+    def lotka_volterra(cls) -> "PolynomialRHS":
+        """The LV field as a table (cross-checks the hand-derived kernel)."""
+        return cls(
+            terms=[
+                (0, 0, 1.0, (1, 0)),   # +alpha * prey
+                (0, 1, -1.0, (1, 1)),  # -beta  * prey * pred
+                (1, 3, 1.0, (1, 1)),   # +delta * prey * pred
+                (1, 2, -1.0, (0, 1)),  # -gamma * pred
+            ],
+            D=2,
+            P=4,
+        )
+
+    @classmethod
+    def sir(cls) -> "PolynomialRHS":
+        """SIR epidemic model: dS=-b S I, dI=b S I - g I, dR=g I."""
+        return cls(
+            terms=[
+                (0, 0, -1.0, (1, 1, 0)),
+                (1, 0, 1.0, (1, 1, 0)),
+                (1, 1, -1.0, (0, 1, 0)),
+                (2, 1, 1.0, (0, 1, 0)),
+            ],
+            D=3,
+            P=2,
+        )
 
 
 def _rk4_step(f, t, u, h, theta):
@@ -106,21 +189,67 @@ class ODEModel(LogpGradModel):
         self._use_kernels = use_kernels
         self._native_state = None  # lazy (obs_of_step, states_ws, out)
 
-    def _native_path(self) -> bool:
-        """Native CDNA4 forward+adjoint kernels (Lotka-Volterra family only):
-        one lane integrates one experiment's whole trajectory in-register."""
+    def _native_kind(self):
+        """Which native CDNA4 forward+adjoint path applies: the hand-derived
+        Lotka-Volterra kernels ("lv"), the coefficient-table polynomial-RHS
+        kernels ("poly", any PolynomialRHS within the table limits), or None
+        (torch eager adjoint sweep)."""
         want = self._u0.is_cuda if self._use_kernels is None else self._use_kernels
-        return (
+        if not (
             want
             and self._u0.is_cuda
-            and self.f is lotka_volterra_rhs
             and self._obs_components is None
             and self._dtype == torch.float64
-            and self._u0.shape[-1] == 2
+        ):
+            return None
+        if self.f is lotka_volterra_rhs and self._u0.shape[-1] == 2:
+            return "lv"
+        if (
+            isinstance(self.f, PolynomialRHS)
+            and self.f.native_ok()
+            and self._u0.shape[-1] == self.f.D
+        ):
+            return "poly"
+        return None
+
+    def _native_path(self) -> bool:
+        return self._native_kind() is not None
+
+    def _poly_native(self, theta_cp: torch.Tensor):
+        """Generic polynomial-RHS native eval: theta_cp [C, P] -> out [C, 1+P]."""
+        import math as _math
+
+        from ..ops import ode_poly_logp_grad
+
+        rhs: PolynomialRHS = self.f
+        B, D = self._u0.shape
+        C = theta_cp.shape[0]
+        if self._native_state is None:
+            obs_of_step = torch.full((self._n_steps + 1,), -1, dtype=torch.int32)
+            for j, idx in enumerate(self._obs_idx):
+                obs_of_step[idx] = j
+            self._native_state = (obs_of_step.to(self._u0.device), None, None)
+        obs_of_step, _, _ = self._native_state
+        ws = getattr(self, "_poly_ws", None)
+        need = C * (self._n_steps + 1) * B * D
+        if ws is None or ws.numel() < need:
+            ws = torch.empty(need, dtype=torch.float64, device=self._u0.device)
+            self._poly_ws = ws
+        out = ode_poly_logp_grad(
+            rhs.terms, D, rhs.P, self._u0, self._y, obs_of_step,
+            self._n_steps, self._h, self._sigma, theta_cp, ws,
         )
+        n_vals = self._y.numel()
+        logp_const = -0.5 * n_vals * _math.log(2.0 * _math.pi * self._sigma**2)
+        return out, logp_const
 
     def _logp_grad_native(self, theta: torch.Tensor):
         import math as _math
+
+        if self._native_kind() == "poly":
+            theta_cp = theta.detach().to(torch.float64).reshape(1, self.f.P)
+            out, logp_const = self._poly_native(theta_cp)
+            return out[0, 0] + logp_const, [out[0, 1:]]
 
         from ..ops import ode_lv_logp_grad
 
@@ -172,10 +301,16 @@ class ODEModel(LogpGradModel):
         import math as _math
 
         theta_c = torch.as_tensor(theta_c)
-        if theta_c.dim() != 2 or theta_c.shape[0] != 4:
-            raise ValueError(f"theta must be [4, C], got {tuple(theta_c.shape)}")
+        n_par = self.f.P if isinstance(self.f, PolynomialRHS) else 4
+        if theta_c.dim() != 2 or theta_c.shape[0] != n_par:
+            raise ValueError(f"theta must be [{n_par}, C], got {tuple(theta_c.shape)}")
         C = theta_c.shape[1]
-        if self._native_path():
+        if self._native_kind() == "poly":
+            out, logp_const = self._poly_native(
+                theta_c.t().contiguous().to(device=self._u0.device, dtype=torch.float64)
+            )
+            return out[:, 0] + logp_const, out[:, 1:].t().contiguous()
+        if self._native_kind() == "lv":
             from ..ops import ode_lv_logp_grad_batched
 
             if self._native_state is None:
@@ -200,7 +335,7 @@ class ODEModel(LogpGradModel):
         for c in range(C):
             logp, (g,) = self.logp_grad(theta_c[:, c])
             logps.append(torch.tensor(float(logp), dtype=torch.float64))
-            grads.append(torch.as_tensor(g, dtype=torch.float64).reshape(4))
+            grads.append(torch.as_tensor(g, dtype=torch.float64).reshape(n_par))
         return torch.stack(logps), torch.stack(grads, dim=1)
 
     def logp_grad(self, theta) -> Tuple[torch.Tensor, List[torch.Tensor]]:

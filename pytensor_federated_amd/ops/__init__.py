@@ -23,6 +23,7 @@ __all__ = [
     "gaussian_linear_logp_grad",
     "gaussian_linear_eval_sync",
     "logistic_glm_logp_grad",
+    "ode_poly_logp_grad",
 ]
 
 _FED_F32, _FED_F64, _FED_BF16 = 0, 1, 2
@@ -98,6 +99,14 @@ def _try_load() -> Optional[ctypes.CDLL]:
     ]
     lib.fed_ode_lv_eval_batched.restype = ctypes.c_int
     lib.fed_ode_lv_eval_batched.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_double, ctypes.c_double,
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+    ]
+    lib.fed_ode_poly_eval.restype = ctypes.c_int
+    lib.fed_ode_poly_eval.argtypes = [
+        ctypes.c_int, ctypes.c_int, ctypes.c_int,
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
         ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_double, ctypes.c_double,
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
@@ -265,6 +274,58 @@ def ode_lv_logp_grad(
         _stream_ptr(),
     )
     _check(rc, "fed_ode_lv_eval")
+    return out
+
+
+PT_MAXD = 4
+
+
+def ode_poly_logp_grad(
+    terms,                     # list of (d, j, coef, exponents[≤4])
+    D: int,
+    P: int,
+    u0: torch.Tensor,          # [B, D] f64
+    y_obs: torch.Tensor,       # [n_obs, B, D] f64
+    obs_of_step: torch.Tensor, # [n_steps+1] int32
+    n_steps: int,
+    h: float,
+    sigma: float,
+    theta: torch.Tensor,       # [C, P] f64 (C chains; C=1 for single eval)
+    states_ws: torch.Tensor,   # [C*(n_steps+1)*B*D] f64 scratch
+    out: Optional[torch.Tensor] = None,  # f64[C, 1+P]
+) -> torch.Tensor:
+    """Generic polynomial-RHS forward+adjoint: out[c] = [logp_quad, g_theta[P]].
+
+    The RHS family is du_d/dt = sum_t c_t * theta_{j_t} * prod_i u_i^{e_ti}
+    interpreted from a term table (see csrc/ode_poly.hip) -- one compiled
+    kernel serves Lotka-Volterra, SIR, oscillators, mass-action kinetics...
+    The logp normalization constant is added by the caller.
+    """
+    lib = require_kernels()
+    B = u0.shape[0]
+    T = len(terms)
+    d_arr = (ctypes.c_int * T)(*[int(t[0]) for t in terms])
+    j_arr = (ctypes.c_int * T)(*[int(t[1]) for t in terms])
+    c_arr = (ctypes.c_double * T)(*[float(t[2]) for t in terms])
+    e_flat = []
+    for t in terms:
+        e = list(t[3]) + [0] * (PT_MAXD - len(t[3]))
+        e_flat.extend(int(v) for v in e[:PT_MAXD])
+    e_arr = (ctypes.c_ubyte * (T * PT_MAXD))(*e_flat)
+    theta_dev = theta.detach().to(device=u0.device, dtype=torch.float64).contiguous()
+    C = theta_dev.shape[0] if theta_dev.dim() == 2 else 1
+    theta_dev = theta_dev.reshape(C, P)
+    if out is None:
+        out = torch.empty((C, 1 + P), dtype=torch.float64, device=u0.device)
+    rc = lib.fed_ode_poly_eval(
+        T, int(D), int(P),
+        d_arr, j_arr, c_arr, e_arr,
+        u0.data_ptr(), y_obs.data_ptr(), obs_of_step.data_ptr(),
+        int(n_steps), int(B), int(C), float(h), float(sigma),
+        theta_dev.data_ptr(), states_ws.data_ptr(), out.data_ptr(),
+        _stream_ptr(),
+    )
+    _check(rc, "fed_ode_poly_eval")
     return out
 
 

@@ -1157,21 +1157,24 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched(
 // 10.8% LDS bank-conflict cycles (phase-B transposed u16 reads: rows r/r+8
 // land on one bank at the 16B-aligned row stride).  v2 changes:
 //
-//  * write-after-barrier staging (T14): the LDS write of chunk c+1 happens
-//    AFTER barrier c and chunk c+2's global loads re-issue immediately, so
-//    every wave gives its loads a full chunk-step of latency instead of
-//    only the MFMA span.
+//  * (a write-after-barrier staging variant was tried and REVERTED: hipcc
+//    coalesced the 4 staging loads into one register quad with a serial
+//    vmcnt(0) per load -- 2.8 ms vs 1.57.  v1's load placement, chunk
+//    c+1's loads issued before chunk c's MFMAs, is what keeps 4 loads in
+//    flight under this compiler.)
 //  * XOR swizzle of the 16-byte group index, key = ((row>>3)&1)<<1 --
 //    rows r and r+8 (the colliding pair within a 32-lane half) get their
 //    group-bit-1 flipped against each other, separating their banks for
 //    the phase-B scalar reads while leaving the b128 write/read patterns
 //    near-conflict-free.
-//  * out-of-range rows handled by address clamping (their logp/resid
-//    contributions are masked later) instead of per-element zero-fill
-//    branches, which de-pipeline hipcc's load scheduling.
+//  * edge rows keep v1's conditional zero-fill: an address-clamp variant
+//    (branch-free) made hipcc spill the staged loads to scratch with a
+//    vmcnt(0) after EACH -- serial HBM round trips (2.8 ms).  With the
+//    conditional form the 4 loads stay in 4 register quads, in flight
+//    together (verified in the ISA).
 
 template <int K>
-__global__ __launch_bounds__(256, 3) void k_logistic_glm_batched_v2(
+__global__ __launch_bounds__(256) void k_logistic_glm_batched_v2(
     const unsigned short* __restrict__ X,   // [N][K] bf16
     const unsigned short* __restrict__ y,   // [N] bf16
     long long n_rows,
@@ -1229,26 +1232,26 @@ __global__ __launch_bounds__(256, 3) void k_logistic_glm_batched_v2(
         U4 ld[4];
         // global loads are linear (coalesced); the XOR swizzle is applied at
         // the LDS WRITE, and compensated at both read sites.
-#define LOAD_CHUNK2(c)                                                             _Pragma("unroll") for (int rr = 0; rr < 4; ++rr) {                                 long long row = row0 + st_r0 + rr * 16;                                        row = row > row_max ? row_max : row;                                           ld[rr] = *(const U4*)&X[row * (long long)K + (c) * BL_CHUNK + st_g * 8];   }
+#define LOAD_CHUNK2(c)                                                             _Pragma("unroll") for (int rr = 0; rr < 4; ++rr) {                                 const long long row = row0 + st_r0 + rr * 16;                                  ld[rr] = (U4){0, 0, 0, 0};                                                     if (row < n_rows)                                                                  ld[rr] = *(const U4*)&X[row * (long long)K + (c) * BL_CHUNK + st_g * 8];   }
 #define WRITE_CHUNK2(buf)                                                          _Pragma("unroll") for (int rr = 0; rr < 4; ++rr)                                   *(U4*)&x_lds[(buf) * x_buf + (st_r0 + rr * 16) * x_stride +                                 (st_g ^ st_key) * 8] = ld[rr];
 
         LOAD_CHUNK2(0)
         WRITE_CHUNK2(0)
-        LOAD_CHUNK2(1)
         int cur = 0;
 
-        // ---- phase A: Z = X . Theta (theta fragments from L2) ----
-        // (not unrolled: full unrolling lets the scheduler hoist every
-        // chunk's loads and blow past the 2-wave/SIMD register budget;
-        // phase B must unroll for static g_acc indexing, this loop not)
+        // ---- phase A: Z = X . Theta ----
+        // v1's load placement is kept deliberately: issuing chunk c+1's
+        // loads BEFORE the MFMAs forces the allocator to keep 4 separate
+        // load register sets alive across them (4 loads in flight).  A
+        // write-after-barrier variant (load c+2 right after the staging
+        // write) let hipcc coalesce all 4 loads into ONE register quad
+        // with s_waitcnt vmcnt(0) after EACH -- serial HBM round trips,
+        // measured 2.81 ms vs 1.57 (see git history).
         f32x4_t z_acc = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll 1
+#pragma unroll
         for (int c = 0; c < n_chunks; ++c) {
             __syncthreads();  // buf[cur] visible (and y_lds on c==0)
-            if (c + 1 < n_chunks) {
-                WRITE_CHUNK2(cur ^ 1)
-                if (c + 2 < n_chunks) LOAD_CHUNK2(c + 2)
-            }
+            if (c + 1 < n_chunks) LOAD_CHUNK2(c + 1)
             const int arow = wid * 16 + (lane & 15);
             const int akey = (((arow >> 3) & 1) << 1);
 #pragma unroll
@@ -1260,6 +1263,7 @@ __global__ __launch_bounds__(256, 3) void k_logistic_glm_batched_v2(
                 b.q = *(U4*)&th_lds[(lane & 15) * th_stride + bk];
                 z_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, z_acc, 0, 0, 0);
             }
+            if (c + 1 < n_chunks) WRITE_CHUNK2(cur ^ 1)
             cur ^= 1;
         }
 
@@ -1286,17 +1290,13 @@ __global__ __launch_bounds__(256, 3) void k_logistic_glm_batched_v2(
         }
         __syncthreads();  // R complete; x_lds free
         WRITE_CHUNK2(0)
-        LOAD_CHUNK2(1)
         cur = 0;
 
         // ---- phase B: G += X_chunk^T . R (chunks L2-hot) ----
 #pragma unroll
         for (int c = 0; c < n_chunks; ++c) {
             __syncthreads();
-            if (c + 1 < n_chunks) {
-                WRITE_CHUNK2(cur ^ 1)
-                if (c + 2 < n_chunks) LOAD_CHUNK2(c + 2)
-            }
+            if (c + 1 < n_chunks) LOAD_CHUNK2(c + 1)
 #pragma unroll
             for (int t2 = 0; t2 < 2; ++t2) {
                 const int kcol0 = wid * 32 + t2 * 16;
@@ -1320,6 +1320,7 @@ __global__ __launch_bounds__(256, 3) void k_logistic_glm_batched_v2(
                 }
                 g_acc[c * 2 + t2] = acc;
             }
+            if (c + 1 < n_chunks) WRITE_CHUNK2(cur ^ 1)
             cur ^= 1;
         }
         __syncthreads();  // rt_lds reuse next tile
