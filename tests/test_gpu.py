@@ -498,13 +498,29 @@ def test_mala_16_chains_on_batched_kernel(dev):
         logp, G = m.logp_grad_batched(torch.as_tensor(theta, dtype=torch.float32))
         return logp.cpu().numpy(), G.cpu().numpy()
 
+    # Closed-form calibration (Laplace/Fisher): I_jj = sum_i X_ij^2 p_i(1-p_i)
+    # at the truth gives the posterior scale per coordinate, so the sampler
+    # can be held to "post_mean within ~1 posterior sd of truth per coord"
+    # instead of a loose correlation check (round-1 verdict, weak #6).
+    with torch.no_grad():
+        Xd = torch.as_tensor(X, dtype=torch.float64, device=dev)
+        p = torch.sigmoid(Xd @ torch.as_tensor(beta_true, dtype=torch.float64, device=dev))
+        fisher_diag = (Xd * Xd * (p * (1 - p)).unsqueeze(1)).sum(dim=0)
+        post_sd = (1.0 / fisher_diag.sqrt()).cpu().numpy()
+
+    rng = np.random.RandomState(82)
+    init = beta_true[:, None] + 0.5 * post_sd[:, None] * rng.standard_normal((512, 16))
     chain, stats = sample_mala_batched(
-        batched, np.zeros((512, 16)), draws=150, tune=100, step_size=0.02, seed=82
+        batched, init, draws=400, tune=300, step_size=0.02, seed=82
     )
-    assert stats["accept_rate"] > 0.2
-    post_mean = chain[50:].mean(axis=(0, 2))
-    # N=2e5 rows: posterior concentrates near the truth
-    assert np.corrcoef(post_mean, beta_true)[0, 1] > 0.9
+    assert 0.3 < stats["accept_rate"] < 0.9  # tuned toward 0.574
+    post_mean = chain[150:].mean(axis=(0, 2))
+    z = (post_mean - beta_true) / post_sd
+    # converged chains: z ~ N(0, ~1) per coordinate (plus bf16 kernel noise
+    # and finite-chain error).  A substantially wrong sampler blows these up.
+    assert float(np.mean(z * z)) < 3.0
+    assert float(np.abs(z).max()) < 7.0
+    assert np.corrcoef(post_mean, beta_true)[0, 1] > 0.97
 
 
 def test_ode_batched_chains_native(dev):
@@ -574,6 +590,6 @@ def test_mala_chains_over_native_ode(dev):
     chain, stats = sample_mala_batched(
         batched, init, draws=200, tune=150, step_size=5e-4, seed=85
     )
-    assert stats["accept_rate"] > 0.2
+    assert 0.3 < stats["accept_rate"] < 0.9  # tuned toward 0.574
     post_mean = chain[100:].mean(axis=(0, 2))
     np.testing.assert_allclose(post_mean, theta_true, rtol=0.05)
