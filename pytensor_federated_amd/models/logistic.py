@@ -190,6 +190,27 @@ class LogisticGLMModel(LogpGradModel):
         X, y = self._X[:, : self._k], self._y
         acc_dtype = torch.float64 if X.dtype == torch.float64 else torch.float32
         theta = theta.to(device=X.device, dtype=acc_dtype)
+        if X.dtype == torch.bfloat16 and X.is_cuda and X.shape[0] > 1 << 20:
+            # large GPU shard: chunk the rows so the f32 upcast of X stays
+            # O(chunk) instead of materializing a full second copy of a
+            # multi-GB shard per call (this is the fallback path for K or B
+            # outside the MFMA kernel's coverage)
+            B = theta.shape[1]
+            logp = torch.zeros(B, dtype=torch.float64, device=X.device)
+            G = torch.zeros((self._k, B), dtype=torch.float64, device=X.device)
+            yf_all = y
+            step = 1 << 21
+            for s in range(0, X.shape[0], step):
+                Xf = X[s : s + step].to(acc_dtype)
+                yf = yf_all[s : s + step].to(acc_dtype)
+                Z = Xf @ theta
+                logp += torch.sum(
+                    yf[:, None] * Z - torch.nn.functional.softplus(Z),
+                    dim=0, dtype=torch.float64,
+                )
+                R = yf[:, None] - torch.sigmoid(Z)
+                G += (Xf.t() @ R).to(torch.float64)
+            return logp, G
         Xf = X.to(acc_dtype)
         yf = y.to(acc_dtype)
         Z = Xf @ theta
