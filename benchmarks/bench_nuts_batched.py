@@ -26,6 +26,7 @@ def main():
     parser.add_argument("--draws", type=int, default=200)
     parser.add_argument("--tune", type=int, default=400)
     parser.add_argument("--mass", default="diag", choices=["diag", "dense"])
+    parser.add_argument("--adaptation", default="simple", choices=["simple", "windowed"])
     args = parser.parse_args()
 
     import torch
@@ -59,7 +60,7 @@ def main():
     t0 = time.perf_counter()
     chain, stats = sample_nuts_batched(
         batched, init, draws=args.draws, tune=args.tune, step_size=5e-4,
-        seed=87, max_depth=8, mass=args.mass,
+        seed=87, max_depth=8, mass=args.mass, adaptation=args.adaptation,
     )
     wall = time.perf_counter() - t0
     post_mean = chain.mean(axis=(0, 2))
@@ -85,7 +86,8 @@ def main():
             split_rhat(chain[:, k, :].T) for k in range(4)
         )),
         "config": {"chains": C, "experiments": args.experiments,
-                   "model": "lotka_volterra_ode_adjoint", "dtype": "f64", "mass": args.mass,
+                   "model": "lotka_volterra_ode_adjoint", "dtype": "f64",
+                   "mass": args.mass, "adaptation": args.adaptation,
                    "kernel": "k_lv_forward_batched/k_lv_adjoint_batched"},
     }))
 

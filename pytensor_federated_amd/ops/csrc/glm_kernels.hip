@@ -1267,8 +1267,8 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v2(
             cur ^= 1;
         }
 
-        // ---- logp + R from Z (prefetch chunk 0 for phase B) ----
-        LOAD_CHUNK2(0)
+        // ---- logp + R from Z (prefetch chunk n-1 for phase B) ----
+        LOAD_CHUNK2(n_chunks - 1)
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
             const int row_in_wave = (lane >> 4) * 4 + r;
@@ -1289,14 +1289,19 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v2(
             rt_lds[chain * rt_stride + row_in_tile] = (unsigned short)((cv.u + rnd) >> 16);
         }
         __syncthreads();  // R complete; x_lds free
-        WRITE_CHUNK2(0)
+        WRITE_CHUNK2(n_chunks - 1)
         cur = 0;
 
-        // ---- phase B: G += X_chunk^T . R (chunks L2-hot) ----
+        // ---- phase B: G += X_chunk^T . R, chunks walked in REVERSE ----
+        // phase A finished on chunk n-1, so that chunk's lines are the
+        // L2-hottest; walking n-1..0 maximizes phase B's L2 hit rate on the
+        // re-read (forward order re-read chunk 0 first, the line most
+        // likely already evicted).
 #pragma unroll
-        for (int c = 0; c < n_chunks; ++c) {
+        for (int ci = 0; ci < n_chunks; ++ci) {
+            const int c = n_chunks - 1 - ci;
             __syncthreads();
-            if (c + 1 < n_chunks) LOAD_CHUNK2(c + 1)
+            if (c - 1 >= 0) LOAD_CHUNK2(c - 1)
 #pragma unroll
             for (int t2 = 0; t2 < 2; ++t2) {
                 const int kcol0 = wid * 32 + t2 * 16;
@@ -1320,7 +1325,7 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v2(
                 }
                 g_acc[c * 2 + t2] = acc;
             }
-            if (c + 1 < n_chunks) WRITE_CHUNK2(cur ^ 1)
+            if (c - 1 >= 0) WRITE_CHUNK2(cur ^ 1)
             cur ^= 1;
         }
         __syncthreads();  // rt_lds reuse next tile
