@@ -1289,7 +1289,7 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v2(
             rt_lds[chain * rt_stride + row_in_tile] = (unsigned short)((cv.u + rnd) >> 16);
         }
         __syncthreads();  // R complete; x_lds free
-        WRITE_CHUNK2(n_chunks - 1)
+        WRITE_CHUNK2(0)   // buffer 0 <- the chunk-(n-1) data loaded above
         cur = 0;
 
         // ---- phase B: G += X_chunk^T . R, chunks walked in REVERSE ----
