@@ -26,7 +26,7 @@ def main():
     parser.add_argument("--draws", type=int, default=200)
     parser.add_argument("--tune", type=int, default=400)
     parser.add_argument("--mass", default="diag", choices=["diag", "dense"])
-    parser.add_argument("--adaptation", default="simple", choices=["simple", "windowed"])
+    parser.add_argument("--adaptation", default="windowed", choices=["simple", "windowed"])
     args = parser.parse_args()
 
     import torch

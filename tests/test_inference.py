@@ -148,7 +148,8 @@ class TestNUTSBatched:
         chain_ref = sample_nuts(single, [np.zeros(2)], draws=60, tune=40, seed=11)
         ref = np.stack([d[0] for d in chain_ref])
         chain, stats = sample_nuts_batched(
-            batched, np.zeros((2, 1)), draws=60, tune=40, seed=11
+            batched, np.zeros((2, 1)), draws=60, tune=40, seed=11,
+            adaptation="simple",  # sequential sample_nuts has no windowed mode
         )
         np.testing.assert_array_equal(chain[:, :, 0], ref)
         assert stats["chains"] == 1
@@ -494,7 +495,8 @@ class TestLockstepEquivalenceProperty:
                                           draws=draws, tune=tune, seed=seed)
             ])
             chain, _ = sample_nuts_batched(
-                batched, init[:, None].copy(), draws=draws, tune=tune, seed=seed
+                batched, init[:, None].copy(), draws=draws, tune=tune, seed=seed,
+                adaptation="simple",  # bit-identity vs the sequential sampler
             )
             np.testing.assert_array_equal(chain[:, :, 0], ref,
                                           err_msg=f"trial {trial}")
