@@ -251,3 +251,62 @@ def test_cpp_worker_serves_grpc(tmp_path):
     finally:
         proc.terminate()
         proc.wait(timeout=10)
+
+
+@pytest.mark.timeout(300)
+def test_cpp_worker_concurrent_clients(tmp_path):
+    """Concurrent clients evaluate DIFFERENT thetas against one worker; each
+    must get its own correct result (per-connection EvalCtx: private stream
+    + buffers -- regression for the round-1 global-eval-mutex design)."""
+    import threading
+
+    from pytensor_federated_amd.common import LogpGradServiceClient
+    from pytensor_federated_amd.models import GaussianLinearModel, generate_linear_dataset
+
+    x, y = generate_linear_dataset(500_000, seed=55)
+    shard = tmp_path / "shard.bin"
+    with open(shard, "wb") as f:
+        f.write(struct.pack("<q", len(x)))
+        f.write(np.asarray(x, dtype=np.float64).tobytes())
+        f.write(np.asarray(y, dtype=np.float64).tobytes())
+
+    env = dict(os.environ, FEDOPS_LIB=str(LIB))
+    port = PORT + 9
+    proc = subprocess.Popen(
+        [str(WORKER), "--port", str(port), "--data", str(shard), "--sigma", "0.4",
+         "--dtype", "bf16"],
+        env=env, stderr=subprocess.PIPE,
+    )
+    try:
+        _wait_tcp(port)
+        import torch
+
+        ref_model = GaussianLinearModel(
+            x, y, sigma=0.4, device="cuda:0", dtype=torch.bfloat16, use_kernels=True
+        )
+        thetas = [(1.5 + 0.1 * i, 0.5 - 0.02 * i) for i in range(4)]
+        refs = [ref_model(a, b) for a, b in thetas]
+        errors = []
+
+        def run_client(idx):
+            try:
+                client = LogpGradServiceClient("127.0.0.1", port, transport="fast")
+                a, b = thetas[idx]
+                for _ in range(50):
+                    logp, (ga, gb) = client.evaluate(a, b)
+                    logp_ref, (ga_ref, gb_ref) = refs[idx]
+                    np.testing.assert_allclose(float(logp), float(logp_ref), rtol=1e-9)
+                    np.testing.assert_allclose(float(ga), float(ga_ref), rtol=1e-7)
+                del client
+            except Exception as ex:  # surface into the main thread
+                errors.append((idx, repr(ex)))
+
+        threads = [threading.Thread(target=run_client, args=(i,)) for i in range(4)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(timeout=120)
+        assert not errors, errors
+    finally:
+        proc.terminate()
+        proc.wait(timeout=10)
