@@ -42,6 +42,23 @@ def _wait_tcp(port, timeout=60.0):
     raise TimeoutError(f"port {port} never opened")
 
 
+def _client_proc(cport, kwargs, n, ready, go):
+    """Top-level (picklable) client driver for the multi-process row.
+
+    Imports + connects, signals ready, then waits for the collective start
+    so process startup stays outside the timed window."""
+    from pytensor_federated_amd.service import ArraysToArraysServiceClient
+
+    client = ArraysToArraysServiceClient("127.0.0.1", cport, **kwargs)
+    a, b = np.float64(1.5), np.float64(0.5)
+    client.evaluate(a, b)  # connect + warm
+    ready.release()
+    go.acquire()
+    for _ in range(n):
+        client.evaluate(a, b)
+    del client
+
+
 def main():
     parser = argparse.ArgumentParser()
     parser.add_argument("--calls", type=int, default=1000)
@@ -50,6 +67,8 @@ def main():
     parser.add_argument("--port", type=int, default=9651)
     parser.add_argument("--echo", action="store_true",
                         help="GPU-less transport-only mode (--model echo)")
+    parser.add_argument("--clients", type=int, default=1,
+                        help="concurrent clients (per edge) for the throughput row")
     args = parser.parse_args()
 
     from pytensor_federated_amd.service import ArraysToArraysServiceClient
@@ -75,6 +94,46 @@ def main():
     try:
         _wait_tcp(gport)
         result = {}
+
+        def one_client(cport, kwargs, n):
+            client = ArraysToArraysServiceClient("127.0.0.1", cport, **kwargs)
+            a, b = np.float64(1.5), np.float64(0.5)
+            for _ in range(n):
+                client.evaluate(a, b)
+            del client
+
+        if args.clients > 1:
+            # concurrent-client throughput: one client PROCESS each (GIL-free
+            # drivers); per-connection EvalCtx on the worker lets their
+            # evaluations overlap on private HIP streams
+            import multiprocessing
+
+            ctx = multiprocessing.get_context("spawn")
+            for name, kwargs in [("grpc_stream", dict(transport="grpc")),
+                                 ("fast", dict(transport="fast"))]:
+                cport = gport if name.startswith("grpc") else port
+                one_client(cport, kwargs, args.warmup)  # server warm
+                ready = ctx.Semaphore(0)
+                go = ctx.Semaphore(0)
+                procs = [
+                    ctx.Process(target=_client_proc,
+                                args=(cport, kwargs, args.calls, ready, go))
+                    for _ in range(args.clients)
+                ]
+                for p in procs:
+                    p.start()
+                for _ in procs:
+                    ready.acquire()  # all imported + connected
+                t0 = time.perf_counter()
+                for _ in procs:
+                    go.release()
+                for p in procs:
+                    p.join()
+                wall = time.perf_counter() - t0
+                result[f"{name}_x{args.clients}"] = {
+                    "aggregate_calls_per_s": args.calls * args.clients / wall,
+                    "us_per_call_per_client": wall / args.calls * 1e6,
+                }
         for name, kwargs in [
             ("grpc_stream", dict(transport="grpc", use_stream=True)),
             ("grpc_unary", dict(transport="grpc", use_stream=False)),
