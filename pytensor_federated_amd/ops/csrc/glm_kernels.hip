@@ -1514,8 +1514,16 @@ extern "C" int fed_logistic_glm_batched(
     const int block = 256;
     const char* v1_env = getenv("FED_BATCHED_V1");
     const bool v1_variant = v1_env && atoi(v1_env) != 0;
-    // v2 runs 3 blocks/CU (~38 KB LDS); 768 blocks covers every CU thrice
-    const int grid_cap = v1_variant ? 304 : 512;  // v2: 2 blocks/CU
+    // v2 runs 2 blocks/CU (71 KB LDS); 512 blocks covers every CU twice.
+    // FED_BATCHED_GRID overrides for the L2-residency A/B: at 256 (1
+    // block/CU) a block's whole 128 KB tile fits its XCD L2 share, so
+    // phase B's re-read can hit L2 instead of HBM -- at the cost of half
+    // the waves hiding phase A's stream latency.
+    int grid_cap = v1_variant ? 304 : 512;
+    if (const char* ge = getenv("FED_BATCHED_GRID")) {
+        int v = atoi(ge);
+        if (v >= 16 && v <= 768) grid_cap = v;
+    }
     int grid = pick_grid(n_rows / BL_ROWS + 1, 1);
     if (grid > grid_cap) grid = grid_cap;
     const long long slab_cols = BCH + (long long)K * BCH;
