@@ -1358,6 +1358,304 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v2(
 }
 
 // ---------------------------------------------------------------------------
+// Batched logistic v3: glds-staged, tile-RESIDENT -- phase B never re-reads
+// ---------------------------------------------------------------------------
+// v2 sits at the 2x-traffic floor (~1.03 ms at 2e6x1024: phase B re-reads X
+// from HBM).  v3 removes the re-read: a 32-row x K tile stays resident in
+// LDS across BOTH phases.  Staging is global_load_lds (one 1-KB DMA per
+// 512-col row half, no VGPR round trip), pipelined one tile ahead through
+// three 32-KB half-buffers with COUNTED vmcnt waits and raw s_barriers (a
+// __syncthreads with a glds in flight drains the whole queue -- guide
+// "Pipelining across barriers").  1 block/CU, contiguous tile ranges.
+//
+// Bank swizzle: the pad-free glds image (row stride 512 u16 == 0 mod 64
+// dwords) would put every row on the same banks, so the SOURCE column
+// group of each lane is XOR-permuted by key(row) = rotl4_by2(row & 15):
+// 4-bit-injective (phase-A b128 slots stay distinct) and key(r) ^
+// key(r+8) == 2 (the phase-B u16 lane-pairs that share a bank mod 8 get
+// split by bit1, while bit0 stays free for the kcol-block bit).
+//
+// Per-wave column split is interleaved across halves (wave w owns cols
+// [w*128, w*128+128) of EACH half) so both phases keep all 4 waves busy
+// on whichever half is resident, and the h1 buffer can be refilled for
+// tile t+1 while phase B finishes tile t's h0 columns.
+
+#define V3_ROWS 32        // rows per tile
+#define V3_HALF 512       // columns per half (K = 1024 only)
+
+__device__ __forceinline__ int v3_key(int row) {
+    const int x = row & 15;
+    return ((x << 2) | (x >> 2)) & 15;  // rotl4 by 2
+}
+
+#define V3_ASM_VMCNT(n) asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory")
+#define V3_BARRIER()                                                    \
+    do {                                                                \
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");              \
+        __builtin_amdgcn_s_barrier();                                   \
+    } while (0)
+
+// stage rows [8w, 8w+8) of one 512-col half: one 1-KB LDS-DMA per row,
+// source columns pre-swizzled per lane.  dst rows are wave-uniform.
+//
+// The DMA is issued by INLINE ASM, not the builtin: hipcc tracks builtin
+// glds in its memory model and bundles `s_waitcnt vmcnt(0)` into the
+// lgkm wait of EVERY later ds_read that might alias -- which drains the
+// cross-tile prefetch at each MFMA (observed in the .s).  The asm form
+// leaves its completion entirely to this kernel's hand-counted
+// V3_ASM_VMCNT waits (the glds queue shares vmcnt, FIFO).
+__device__ __forceinline__ void v3_glds_row(const void* src, unsigned lds_byte_off) {
+    asm volatile(
+        "s_mov_b32 m0, %0\n\t"
+        "s_nop 0\n\t"
+        "global_load_lds_dwordx4 %1, off"
+        // no "memory" clobber: it would make hipcc bundle a conservative
+        // vmcnt(0) into every later ds_read's wait (the exact drain this
+        // asm form exists to avoid).  Ordering is carried by the volatile
+        // asm barriers (which DO clobber memory) on either side.
+        :: "s"(lds_byte_off), "v"(src));
+}
+
+__device__ __forceinline__ void v3_stage_half(
+    const unsigned short* __restrict__ X, long long n_rows, int K,
+    long long row0, int colbase, unsigned short* half_buf,
+    const char* smem_base, int wid, int lane
+) {
+#pragma unroll
+    for (int rr = 0; rr < 8; ++rr) {
+        const int r = wid * 8 + rr;
+        const long long grow = row0 + r;
+        unsigned short* dst = half_buf + r * V3_HALF;
+        if (grow < n_rows) {
+            const unsigned short* src =
+                &X[grow * (long long)K + colbase + ((lane ^ v3_key(r)) * 8)];
+            // the row base is wave-uniform by construction; readfirstlane
+            // makes that provable so the asm "s" constraint gets an SGPR
+            v3_glds_row(src, __builtin_amdgcn_readfirstlane(
+                                 (unsigned)((const char*)dst - smem_base)));
+        } else {
+            *(U4*)&dst[lane * 8] = (U4){0, 0, 0, 0};
+        }
+    }
+}
+
+__global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
+    const unsigned short* __restrict__ X,   // [N][1024] bf16
+    const unsigned short* __restrict__ y,   // [N] bf16
+    long long n_rows,
+    const unsigned short* __restrict__ theta_t,  // [16][1024] bf16
+    float* __restrict__ slab                     // [grid][16 + 1024*16]
+) {
+    constexpr int K = 2 * V3_HALF;
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    unsigned short* th_lds = (unsigned short*)smem;        // [16][K+TPAD]
+    const int th_stride = K + TPAD;
+    // three half buffers live at x_base + {0,1,2}*HBUF; they are addressed
+    // by OFFSET so every access provably derives from the __shared__ base
+    // (a pointer array indexed at runtime loses the address-space
+    // inference and hipcc emits flat_load for what should be ds_read)
+    constexpr int HBUF = V3_ROWS * V3_HALF;
+    unsigned short* x_base = th_lds + BCH * th_stride;     // 3 x [32][512]
+    unsigned short* rt_lds = x_base + 3 * HBUF;            // [16][32+RPAD]
+    const int rt_stride = V3_ROWS + RPAD;
+    float* zc_lds = (float*)(rt_lds + BCH * rt_stride);    // [4][32][16]
+    float* red_lds = zc_lds + 4 * V3_ROWS * BCH;           // [256]
+
+    typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
+    typedef __attribute__((ext_vector_type(4))) float f32x4_t;
+    union frag_u { bf16x8_t v; unsigned short u[8]; U4 q; };
+
+    // ---- stage Theta^T once (ordinary loads; BEFORE any glds) ----
+    for (int idx = threadIdx.x * 8; idx < BCH * K; idx += 256 * 8) {
+        const int b = idx / K;
+        const int k = idx % K;
+        *(U4*)&th_lds[b * th_stride + k] = *(const U4*)&theta_t[b * K + k];
+    }
+    __syncthreads();
+
+    f32x4_t g_acc[16];
+#pragma unroll
+    for (int t = 0; t < 16; ++t) g_acc[t] = (f32x4_t){0.f, 0.f, 0.f, 0.f};
+    float logp0 = 0.f, logp1 = 0.f;  // this thread's two (row, chain) slots
+
+    // contiguous tile range per block (L2 locality + equal glds ledgers)
+    const long long n_tiles = (n_rows + V3_ROWS - 1) / V3_ROWS;
+    const long long tpb = (n_tiles + gridDim.x - 1) / gridDim.x;
+    const long long t_begin = blockIdx.x * tpb;
+    const long long t_end = t_begin + tpb < n_tiles ? t_begin + tpb : n_tiles;
+
+    if (t_begin < t_end) {
+        // prologue: stage tile t_begin fully.  h0 buffers alternate 0/2,
+        // h1 lives in buffer 1.
+        v3_stage_half(X, n_rows, K, t_begin * V3_ROWS, 0, x_base, smem, wid, lane);
+        v3_stage_half(X, n_rows, K, t_begin * V3_ROWS, V3_HALF, x_base + HBUF, smem, wid, lane);
+
+        int h0sel = 0;  // buffer index (0 or 2) holding the CURRENT tile's h0
+#pragma unroll 1
+        for (long long tile = t_begin; tile < t_end; ++tile) {
+            const long long row0 = tile * V3_ROWS;
+            unsigned short* h0 = x_base + h0sel * HBUF;
+            unsigned short* h1 = x_base + HBUF;
+            unsigned short* h0n = x_base + (h0sel ^ 2) * HBUF;  // next tile's h0
+            const bool more = tile + 1 < t_end;
+
+            // [1] own h0 DMAs done (h1's 8 may stay in flight), all waves
+            V3_ASM_VMCNT(8);
+            V3_BARRIER();
+            // [2] prefetch next tile's h0 as deep as possible
+            if (more)
+                v3_stage_half(X, n_rows, K, row0 + V3_ROWS, 0, h0n, smem, wid, lane);
+
+            // ---- phase A on h0: z += X[:, w*128 .. +128) . theta ----
+            f32x4_t z0 = {0.f, 0.f, 0.f, 0.f}, z1 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+            for (int ks = 0; ks < 4; ++ks) {
+                const int kc = wid * 128 + ks * 32;  // column in half
+                frag_u b0f, b1f, a0, a1;
+                const int g = (kc >> 3) + (lane >> 4);  // 16B group in half
+                const int arow0 = lane & 15;            // rows 0..15
+                const int arow1 = 16 + (lane & 15);     // rows 16..31
+                a0.q = *(U4*)&h0[arow0 * V3_HALF + ((g ^ v3_key(arow0)) * 8)];
+                a1.q = *(U4*)&h0[arow1 * V3_HALF + ((g ^ v3_key(arow1)) * 8)];
+                const int bk = kc + (lane >> 4) * 8;    // theta col (h0)
+                b0f.q = *(U4*)&th_lds[(lane & 15) * th_stride + bk];
+                z0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0.v, b0f.v, z0, 0, 0, 0);
+                b1f.q = b0f.q;
+                z1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1.v, b1f.v, z1, 0, 0, 0);
+            }
+            // [4] own h1 DMAs done (next-h0's 8 may stay in flight)
+            V3_ASM_VMCNT(8);
+            V3_BARRIER();
+            // ---- phase A on h1 ----
+#pragma unroll
+            for (int ks = 0; ks < 4; ++ks) {
+                const int kc = wid * 128 + ks * 32;
+                frag_u bf, a0, a1;
+                const int g = (kc >> 3) + (lane >> 4);
+                const int arow0 = lane & 15;
+                const int arow1 = 16 + (lane & 15);
+                a0.q = *(U4*)&h1[arow0 * V3_HALF + ((g ^ v3_key(arow0)) * 8)];
+                a1.q = *(U4*)&h1[arow1 * V3_HALF + ((g ^ v3_key(arow1)) * 8)];
+                const int bk = V3_HALF + kc + (lane >> 4) * 8;  // theta col (h1)
+                bf.q = *(U4*)&th_lds[(lane & 15) * th_stride + bk];
+                z0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0.v, bf.v, z0, 0, 0, 0);
+                z1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1.v, bf.v, z1, 0, 0, 0);
+            }
+            // ---- combine the 4 waves' z quarters; logp + R ----
+            {
+                float* zc = zc_lds + wid * V3_ROWS * BCH;
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    zc[((lane >> 4) * 4 + r) * BCH + (lane & 15)] = z0[r];
+                    zc[(16 + (lane >> 4) * 4 + r) * BCH + (lane & 15)] = z1[r];
+                }
+            }
+            V3_BARRIER();
+            {
+#pragma unroll
+                for (int s = 0; s < 2; ++s) {
+                    const int slot = threadIdx.x + s * 256;
+                    const int row = slot >> 4;      // 0..31
+                    const int chain = slot & 15;
+                    float z = 0.f;
+#pragma unroll
+                    for (int w = 0; w < 4; ++w)
+                        z += zc_lds[(w * V3_ROWS + row) * BCH + chain];
+                    const long long grow = row0 + row;
+                    float resid = 0.f;
+                    if (grow < n_rows) {
+                        // ordinary load; the vmcnt(0) it forces only waits
+                        // the next-h0 prefetch (~already landed)
+                        const float yv = bf16_bits_to_f32(y[grow]);
+                        const float sp = fmaxf(z, 0.f) + log1pf(__expf(-fabsf(z)));
+                        const float term = yv * z - sp;
+                        if (s == 0) logp0 += term; else logp1 += term;
+                        resid = yv - 1.f / (1.f + __expf(-z));
+                    }
+                    union { float f; unsigned int u; } cv;
+                    cv.f = resid;
+                    const unsigned int rnd = 0x7fff + ((cv.u >> 16) & 1);
+                    rt_lds[chain * rt_stride + row] =
+                        (unsigned short)((cv.u + rnd) >> 16);
+                }
+            }
+            V3_BARRIER();  // R visible
+
+            // ---- phase B, h1 columns first (frees h1 for the refill) ----
+#pragma unroll
+            for (int t2 = 0; t2 < 8; ++t2) {
+                const int kc = wid * 128 + t2 * 16;       // column in half
+                const int kcol = kc + (lane & 15);
+                f32x4_t acc = g_acc[8 + t2];
+                frag_u a, b;
+#pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    const int row = (lane >> 4) * 8 + j;
+                    a.u[j] = h1[row * V3_HALF +
+                                (((kcol >> 3) ^ v3_key(row)) * 8) + (kcol & 7)];
+                }
+                b.q = *(U4*)&rt_lds[(lane & 15) * rt_stride + (lane >> 4) * 8];
+                g_acc[8 + t2] =
+                    __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, acc, 0, 0, 0);
+            }
+            V3_BARRIER();  // everyone done reading h1
+            // [6] refill h1 with the NEXT tile's second half
+            if (more)
+                v3_stage_half(X, n_rows, K, row0 + V3_ROWS, V3_HALF, h1, smem, wid, lane);
+
+            // ---- phase B, h0 columns ----
+#pragma unroll
+            for (int t2 = 0; t2 < 8; ++t2) {
+                const int kc = wid * 128 + t2 * 16;
+                const int kcol = kc + (lane & 15);
+                f32x4_t acc = g_acc[t2];
+                frag_u a, b;
+#pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    const int row = (lane >> 4) * 8 + j;
+                    a.u[j] = h0[row * V3_HALF +
+                                (((kcol >> 3) ^ v3_key(row)) * 8) + (kcol & 7)];
+                }
+                b.q = *(U4*)&rt_lds[(lane & 15) * rt_stride + (lane >> 4) * 8];
+                g_acc[t2] =
+                    __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, acc, 0, 0, 0);
+            }
+            h0sel ^= 2;
+        }
+    }
+
+    // ---- epilogue: block partials -> slab (layout shared with v1/v2) ----
+    __syncthreads();
+    red_lds[threadIdx.x] = logp0 + logp1;
+    // the two slots of one thread are chains (t&15) and ((t+256)&15) == same
+    // chain, different rows -- so the plain per-chain strided sum works
+    __syncthreads();
+    float* slab_blk = slab + (long long)blockIdx.x * (BCH + (long long)K * BCH);
+    if (threadIdx.x < BCH) {
+        float s = 0.f;
+        for (int i = threadIdx.x; i < 256; i += BCH) s += red_lds[i];
+        slab_blk[threadIdx.x] = s;
+    }
+    float* g_slab = slab_blk + BCH;
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+#pragma unroll
+        for (int t2 = 0; t2 < 8; ++t2) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int kcol = half * V3_HALF + wid * 128 + t2 * 16 +
+                                 (lane >> 4) * 4 + r;
+                const int chain = lane & 15;
+                g_slab[(long long)kcol * BCH + chain] = g_acc[half * 8 + t2][r];
+            }
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
 // Batched logistic, LDS-resident-tile variant: phase B never re-reads HBM
 // ---------------------------------------------------------------------------
 // The chunked variant's phase-B re-read misses L2 (2 blocks/CU x 32 CUs x
@@ -1538,6 +1836,28 @@ extern "C" int fed_logistic_glm_batched(
         (BL_ROWS + 256) * 4 + 64;
     const char* lds_env = getenv("FED_BATCHED_LDS");
     const bool lds_variant = lds_env && atoi(lds_env) != 0;
+    const char* v3_env = getenv("FED_BATCHED_V3");
+    if (v3_env && atoi(v3_env) != 0 && K == 1024) {
+        // glds tile-resident variant: 1 block/CU, contiguous tile ranges
+        int g3 = grid;
+        if (g3 > 256) g3 = 256;
+        const int lds3 = (BCH * (K + TPAD) + 3 * V3_ROWS * V3_HALF +
+                          BCH * (V3_ROWS + RPAD)) * 2 +
+                         (4 * V3_ROWS * BCH + 256) * 4 + 64;
+        hipLaunchKernelGGL(k_logistic_glm_batched_v3, dim3(g3), dim3(block), lds3,
+                           stream, (const unsigned short*)X, (const unsigned short*)y,
+                           n_rows, (const unsigned short*)theta_t_bf16, workspace);
+        hipError_t verr = hipGetLastError();
+        if (verr != hipSuccess) return (int)verr;
+        const int rg = ((int)slab_cols + 255) / 256;
+        int ch3 = g3 / 8;
+        if (ch3 < 1) ch3 = 1;
+        hipError_t m3 = hipMemsetAsync(out, 0, slab_cols * 8, stream);
+        if (m3 != hipSuccess) return (int)m3;
+        hipLaunchKernelGGL(k_colsum_reduce, dim3(rg, ch3), dim3(256), 0, stream,
+                           workspace, g3, (int)slab_cols, out);
+        return (int)hipGetLastError();
+    }
     if (!lds_variant && !v1_variant) {
         if (K == 1024)
             hipLaunchKernelGGL(k_logistic_glm_batched_v2<1024>, dim3(grid), dim3(block),
