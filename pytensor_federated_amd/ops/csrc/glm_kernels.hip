@@ -1416,6 +1416,20 @@ __device__ __forceinline__ void v3_glds_row(const void* src, unsigned lds_byte_o
         :: "s"(lds_byte_off), "v"(src));
 }
 
+// stage the tile's 32 y values (64 bytes) with ONE 1-byte-per-lane DMA;
+// issued by EVERY wave to the same LDS slot (idempotent, keeps the
+// per-wave vmcnt ledgers identical).  Only for full tiles -- the final
+// partial tile falls back to guarded ordinary loads at the R step.
+__device__ __forceinline__ void v3_glds_y(const unsigned short* y, long long row0,
+                                          unsigned lds_byte_off, int lane) {
+    const char* src = (const char*)(y + row0) + lane;
+    asm volatile(
+        "s_mov_b32 m0, %0\n\t"
+        "s_nop 0\n\t"
+        "global_load_lds_ubyte %1, off"
+        :: "s"(lds_byte_off), "v"(src));
+}
+
 __device__ __forceinline__ void v3_stage_half(
     const unsigned short* __restrict__ X, long long n_rows, int K,
     long long row0, int colbase, unsigned short* half_buf,
@@ -1461,7 +1475,8 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
     unsigned short* x_base = th_lds + BCH * th_stride;     // 3 x [32][512]
     unsigned short* rt_lds = x_base + 3 * HBUF;            // [16][32+RPAD]
     const int rt_stride = V3_ROWS + RPAD;
-    float* zc_lds = (float*)(rt_lds + BCH * rt_stride);    // [4][32][16]
+    unsigned short* y_lds = rt_lds + BCH * rt_stride;      // [32] (DMA slot)
+    float* zc_lds = (float*)(y_lds + V3_ROWS);             // [4][32][16]
     float* red_lds = zc_lds + 4 * V3_ROWS * BCH;           // [256]
 
     typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
@@ -1490,8 +1505,12 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
     if (t_begin < t_end) {
         // prologue: stage tile t_begin fully.  h0 buffers alternate 0/2,
         // h1 lives in buffer 1.
+        const unsigned y_off = __builtin_amdgcn_readfirstlane(
+            (unsigned)((const char*)y_lds - smem));
         v3_stage_half(X, n_rows, K, t_begin * V3_ROWS, 0, x_base, smem, wid, lane);
         v3_stage_half(X, n_rows, K, t_begin * V3_ROWS, V3_HALF, x_base + HBUF, smem, wid, lane);
+        const bool first_full = t_begin * V3_ROWS + V3_ROWS <= n_rows;
+        if (first_full) v3_glds_y(y, t_begin * V3_ROWS, y_off, lane);
 
         int h0sel = 0;  // buffer index (0 or 2) holding the CURRENT tile's h0
 #pragma unroll 1
@@ -1502,8 +1521,10 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
             unsigned short* h0n = x_base + (h0sel ^ 2) * HBUF;  // next tile's h0
             const bool more = tile + 1 < t_end;
 
-            // [1] own h0 DMAs done (h1's 8 may stay in flight), all waves
-            V3_ASM_VMCNT(8);
+            const bool full = row0 + V3_ROWS <= n_rows;
+            // [1] own h0 DMAs done; h1's 8 (+1 y, on full tiles) may stay
+            // in flight, all waves
+            if (full) V3_ASM_VMCNT(9); else V3_ASM_VMCNT(8);
             V3_BARRIER();
             // [2] prefetch next tile's h0 as deep as possible
             if (more)
@@ -1526,7 +1547,7 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
                 b1f.q = b0f.q;
                 z1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1.v, b1f.v, z1, 0, 0, 0);
             }
-            // [4] own h1 DMAs done (next-h0's 8 may stay in flight)
+            // [4] own h1 (+y) DMAs done (next-h0's 8 may stay in flight)
             V3_ASM_VMCNT(8);
             V3_BARRIER();
             // ---- phase A on h1 ----
@@ -1567,9 +1588,12 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
                     const long long grow = row0 + row;
                     float resid = 0.f;
                     if (grow < n_rows) {
-                        // ordinary load; the vmcnt(0) it forces only waits
-                        // the next-h0 prefetch (~already landed)
-                        const float yv = bf16_bits_to_f32(y[grow]);
+                        // full tiles read y from the DMA slot (no tracked
+                        // global load -> hipcc inserts no vmcnt(0) drain in
+                        // the loop); only the final partial tile pays an
+                        // ordinary load
+                        const float yv = full ? bf16_bits_to_f32(y_lds[row])
+                                              : bf16_bits_to_f32(y[grow]);
                         const float sp = fmaxf(z, 0.f) + log1pf(__expf(-fabsf(z)));
                         const float term = yv * z - sp;
                         if (s == 0) logp0 += term; else logp1 += term;
@@ -1601,10 +1625,14 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
                 g_acc[8 + t2] =
                     __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, acc, 0, 0, 0);
             }
-            V3_BARRIER();  // everyone done reading h1
-            // [6] refill h1 with the NEXT tile's second half
-            if (more)
+            V3_BARRIER();  // everyone done reading h1 (and y_lds)
+            // [6] refill h1 (+ the next tile's y) -- the y DMA is issued
+            // right after h1's so the [1]/[4] ledgers stay exact
+            if (more) {
                 v3_stage_half(X, n_rows, K, row0 + V3_ROWS, V3_HALF, h1, smem, wid, lane);
+                if (row0 + 2 * V3_ROWS <= n_rows)
+                    v3_glds_y(y, row0 + V3_ROWS, y_off, lane);
+            }
 
             // ---- phase B, h0 columns ----
 #pragma unroll
