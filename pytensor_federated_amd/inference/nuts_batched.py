@@ -264,7 +264,7 @@ def sample_nuts_batched(
         "windowed" is Stan-style expanding metric windows with a step-size
         restart at each boundary -- measured 2193 vs 1680 draws/s at equal
         split-R-hat (1.006 vs 1.008) on the 16-chain dense-metric LV
-        posterior (gpurun_out/r2c4_nuts_*.json), so it is the default from
+        posterior (profiles/raw_r2/r2c4_nuts_*.json), so it is the default from
         round 2; "simple" keeps the single mid-tune metric update
         "simple" = one metric update at 60% of tune (matches
         ``sample_nuts``, keeps C=1 bit-identity).  "windowed" = Stan-style
