@@ -1416,18 +1416,19 @@ __device__ __forceinline__ void v3_glds_row(const void* src, unsigned lds_byte_o
         :: "s"(lds_byte_off), "v"(src));
 }
 
-// stage the tile's 32 y values (64 bytes) with ONE 1-byte-per-lane DMA;
-// issued by EVERY wave to the same LDS slot (idempotent, keeps the
-// per-wave vmcnt ledgers identical).  Only for full tiles -- the final
-// partial tile falls back to guarded ordinary loads at the R step.
-__device__ __forceinline__ void v3_glds_y(const unsigned short* y, long long row0,
-                                          unsigned lds_byte_off, int lane) {
-    const char* src = (const char*)(y + row0) + lane;
-    asm volatile(
-        "s_mov_b32 m0, %0\n\t"
-        "s_nop 0\n\t"
-        "global_load_lds_ubyte %1, off"
-        :: "s"(lds_byte_off), "v"(src));
+// load one y value through a HIDDEN asm global load: it rides the same
+// vmcnt queue as the tile DMAs (so the [4] counted wait covers it) and,
+// unlike an ordinary tracked load, does not make hipcc insert a
+// conservative vmcnt(0) drain at the use.  OOB rows clamp the address
+// (uniform issue keeps the per-wave ledger exact); the caller masks the
+// value.  (A 1-byte-per-lane LDS-DMA variant of this was tried first and
+// produced wrong y values -- ubyte DMA lane addressing did not match the
+// lane*size model; reverted to register loads.)
+__device__ __forceinline__ unsigned v3_load_y_asm(const unsigned short* addr) {
+    unsigned v;
+    asm volatile("global_load_ushort %0, %1, off"
+                 : "=v"(v) : "v"(addr));
+    return v;
 }
 
 __device__ __forceinline__ void v3_stage_half(
@@ -1475,8 +1476,7 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
     unsigned short* x_base = th_lds + BCH * th_stride;     // 3 x [32][512]
     unsigned short* rt_lds = x_base + 3 * HBUF;            // [16][32+RPAD]
     const int rt_stride = V3_ROWS + RPAD;
-    unsigned short* y_lds = rt_lds + BCH * rt_stride;      // [32] (DMA slot)
-    float* zc_lds = (float*)(y_lds + V3_ROWS);             // [4][32][16]
+    float* zc_lds = (float*)(rt_lds + BCH * rt_stride);    // [4][32][16]
     float* red_lds = zc_lds + 4 * V3_ROWS * BCH;           // [256]
 
     typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
@@ -1505,12 +1505,8 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
     if (t_begin < t_end) {
         // prologue: stage tile t_begin fully.  h0 buffers alternate 0/2,
         // h1 lives in buffer 1.
-        const unsigned y_off = __builtin_amdgcn_readfirstlane(
-            (unsigned)((const char*)y_lds - smem));
         v3_stage_half(X, n_rows, K, t_begin * V3_ROWS, 0, x_base, smem, wid, lane);
         v3_stage_half(X, n_rows, K, t_begin * V3_ROWS, V3_HALF, x_base + HBUF, smem, wid, lane);
-        const bool first_full = t_begin * V3_ROWS + V3_ROWS <= n_rows;
-        if (first_full) v3_glds_y(y, t_begin * V3_ROWS, y_off, lane);
 
         int h0sel = 0;  // buffer index (0 or 2) holding the CURRENT tile's h0
 #pragma unroll 1
@@ -1521,11 +1517,18 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
             unsigned short* h0n = x_base + (h0sel ^ 2) * HBUF;  // next tile's h0
             const bool more = tile + 1 < t_end;
 
-            const bool full = row0 + V3_ROWS <= n_rows;
-            // [1] own h0 DMAs done; h1's 8 (+1 y, on full tiles) may stay
-            // in flight, all waves
-            if (full) V3_ASM_VMCNT(9); else V3_ASM_VMCNT(8);
+            // [1] own h0 DMAs done (h1's 8 may stay in flight), all waves
+            V3_ASM_VMCNT(8);
             V3_BARRIER();
+            // this thread's two y values (R-step slots), hidden loads on
+            // the same queue, issued BEFORE the h0 prefetch so the [4]
+            // counted wait (which leaves only the prefetch outstanding)
+            // provably covers them
+            const long long ymax = n_rows - 1;
+            long long yr0 = row0 + (threadIdx.x >> 4);
+            long long yr1 = row0 + 16 + (threadIdx.x >> 4);
+            const unsigned yb0 = v3_load_y_asm(y + (yr0 > ymax ? ymax : yr0));
+            const unsigned yb1 = v3_load_y_asm(y + (yr1 > ymax ? ymax : yr1));
             // [2] prefetch next tile's h0 as deep as possible
             if (more)
                 v3_stage_half(X, n_rows, K, row0 + V3_ROWS, 0, h0n, smem, wid, lane);
@@ -1547,8 +1550,11 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
                 b1f.q = b0f.q;
                 z1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1.v, b1f.v, z1, 0, 0, 0);
             }
-            // [4] own h1 (+y) DMAs done (next-h0's 8 may stay in flight)
-            V3_ASM_VMCNT(8);
+            // [4] own h1 DMAs + y loads done.  The count must match what
+            // was actually issued after them: 8 prefetch DMAs on interior
+            // tiles, NOTHING on the last tile (an unconditional vmcnt(8)
+            // there would leave h1/y unwaited -- a timing-dependent race).
+            if (more) V3_ASM_VMCNT(8); else V3_ASM_VMCNT(0);
             V3_BARRIER();
             // ---- phase A on h1 ----
 #pragma unroll
@@ -1588,12 +1594,8 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
                     const long long grow = row0 + row;
                     float resid = 0.f;
                     if (grow < n_rows) {
-                        // full tiles read y from the DMA slot (no tracked
-                        // global load -> hipcc inserts no vmcnt(0) drain in
-                        // the loop); only the final partial tile pays an
-                        // ordinary load
-                        const float yv = full ? bf16_bits_to_f32(y_lds[row])
-                                              : bf16_bits_to_f32(y[grow]);
+                        const float yv = bf16_bits_to_f32(
+                            (unsigned short)(s == 0 ? yb0 : yb1));
                         const float sp = fmaxf(z, 0.f) + log1pf(__expf(-fabsf(z)));
                         const float term = yv * z - sp;
                         if (s == 0) logp0 += term; else logp1 += term;
@@ -1625,14 +1627,10 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
                 g_acc[8 + t2] =
                     __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, acc, 0, 0, 0);
             }
-            V3_BARRIER();  // everyone done reading h1 (and y_lds)
-            // [6] refill h1 (+ the next tile's y) -- the y DMA is issued
-            // right after h1's so the [1]/[4] ledgers stay exact
-            if (more) {
+            V3_BARRIER();  // everyone done reading h1
+            // [6] refill h1 with the NEXT tile's second half
+            if (more)
                 v3_stage_half(X, n_rows, K, row0 + V3_ROWS, V3_HALF, h1, smem, wid, lane);
-                if (row0 + 2 * V3_ROWS <= n_rows)
-                    v3_glds_y(y, row0 + V3_ROWS, y_off, lane);
-            }
 
             // ---- phase B, h0 columns ----
 #pragma unroll
