@@ -1862,8 +1862,13 @@ extern "C" int fed_logistic_glm_batched(
         (BL_ROWS + 256) * 4 + 64;
     const char* lds_env = getenv("FED_BATCHED_LDS");
     const bool lds_variant = lds_env && atoi(lds_env) != 0;
+    // v3 (glds tile-resident, 1x HBM traffic) is the default at K=1024:
+    // 0.878 ms vs v2's 1.035 / v1's 1.568 at 2e6x1024x16, 5.17 ms at the
+    // config-4 shard (profiles/raw_r2/r2c10_*).  FED_BATCHED_V3=0 forces
+    // the v2 chunked kernel (which also serves K=512).
     const char* v3_env = getenv("FED_BATCHED_V3");
-    if (v3_env && atoi(v3_env) != 0 && K == 1024) {
+    const bool v3_on = v3_env ? atoi(v3_env) != 0 : true;
+    if (v3_on && !v1_variant && !lds_variant && K == 1024) {
         // glds tile-resident variant: 1 block/CU, contiguous tile ranges
         int g3 = grid;
         if (g3 > 256) g3 = 256;
