@@ -374,7 +374,13 @@ class ODEModel(LogpGradModel):
             theta_k = theta.detach().requires_grad_(True)
             with torch.enable_grad():
                 u_next = _rk4_step(self.f, t, u_k, self._h, theta_k)
-                gu, gth = torch.autograd.grad(u_next, (u_k, theta_k), grad_outputs=lam)
+                gu, gth = torch.autograd.grad(
+                    u_next, (u_k, theta_k), grad_outputs=lam, allow_unused=True
+                )
+            if gu is None:  # constant field: no state dependence
+                gu = torch.zeros_like(u_k)
+            if gth is None:  # theta-free field (e.g. fixed dynamics)
+                gth = torch.zeros_like(theta_k)
             g_theta = g_theta + gth
             lam = gu
             if k in obs_grad:
