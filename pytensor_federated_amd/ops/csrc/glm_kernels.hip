@@ -1454,6 +1454,199 @@ __device__ __forceinline__ void v3_stage_half(
     }
 }
 
+// Small-K sibling of v3 (K <= 512, multiple of 128): a whole 32-row x K
+// tile is ONE <=32-KB buffer, so the pipeline is a plain 3-buffer rotation
+// with prefetch depth TWO tiles (the K=1024 kernel must split tiles into
+// column halves to fit; here each buffer is self-sufficient, which also
+// lets phase A finish Z in one sweep).  Staging, swizzle, hidden y loads
+// and the counted-vmcnt discipline are identical to v3.
+template <int K>
+__global__ __launch_bounds__(256) void k_logistic_glm_batched_v3s(
+    const unsigned short* __restrict__ X,   // [N][K] bf16
+    const unsigned short* __restrict__ y,   // [N] bf16
+    long long n_rows,
+    const unsigned short* __restrict__ theta_t,  // [16][K] bf16
+    float* __restrict__ slab                     // [grid][16 + K*16]
+) {
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    constexpr int QC = K / 4;        // columns per wave quarter
+    constexpr int NKS = QC / 32;     // phase-A MFMA steps per quarter
+    constexpr int NT2 = QC / 16;     // phase-B column tiles per quarter
+    constexpr int TBUF = V3_ROWS * K;  // u16 per tile buffer
+
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    unsigned short* th_lds = (unsigned short*)smem;        // [16][K+TPAD]
+    const int th_stride = K + TPAD;
+    unsigned short* x_base = th_lds + BCH * th_stride;     // 3 x [32][K]
+    unsigned short* rt_lds = x_base + 3 * TBUF;            // [16][32+RPAD]
+    const int rt_stride = V3_ROWS + RPAD;
+    float* zc_lds = (float*)(rt_lds + BCH * rt_stride);    // [4][32][16]
+    float* red_lds = zc_lds + 4 * V3_ROWS * BCH;           // [256]
+
+    typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
+    typedef __attribute__((ext_vector_type(4))) float f32x4_t;
+    union frag_u { bf16x8_t v; unsigned short u[8]; U4 q; };
+
+    for (int idx = threadIdx.x * 8; idx < BCH * K; idx += 256 * 8) {
+        const int b = idx / K;
+        const int k = idx % K;
+        *(U4*)&th_lds[b * th_stride + k] = *(const U4*)&theta_t[b * K + k];
+    }
+    __syncthreads();
+
+    f32x4_t g_acc[NT2];
+#pragma unroll
+    for (int t = 0; t < NT2; ++t) g_acc[t] = (f32x4_t){0.f, 0.f, 0.f, 0.f};
+    float logp0 = 0.f, logp1 = 0.f;
+
+    const long long n_tiles = (n_rows + V3_ROWS - 1) / V3_ROWS;
+    const long long tpb = (n_tiles + gridDim.x - 1) / gridDim.x;
+    const long long t_begin = blockIdx.x * tpb;
+    const long long t_end = t_begin + tpb < n_tiles ? t_begin + tpb : n_tiles;
+
+    // stage one whole tile into buffer `buf`: at K=512 one row is exactly
+    // the 1-KB glds span (row stride = K*2 = 1024 B), so the staging shape
+    // is identical to v3's; fully out-of-range rows zero-fill.
+    static_assert(K == 512, "v3s is instantiated for the padded K=512 granule");
+    auto stage_tile = [&](long long row0, int buf) {
+#pragma unroll
+        for (int rr = 0; rr < 8; ++rr) {
+            const int r = wid * 8 + rr;
+            const long long grow = row0 + r;
+            unsigned short* dst = x_base + buf * TBUF + r * K;
+            if (grow < n_rows) {
+                const unsigned short* src =
+                    &X[grow * (long long)K + ((lane ^ v3_key(r)) * 8)];
+                v3_glds_row(src, __builtin_amdgcn_readfirstlane(
+                                     (unsigned)((const char*)dst - smem)));
+            } else {
+                *(U4*)&dst[lane * 8] = (U4){0, 0, 0, 0};
+            }
+        }
+    };
+
+    if (t_begin < t_end) {
+        stage_tile(t_begin * V3_ROWS, 0);
+        if (t_begin + 1 < t_end) stage_tile((t_begin + 1) * V3_ROWS, 1);
+
+#pragma unroll 1
+        for (long long tile = t_begin; tile < t_end; ++tile) {
+            const long long row0 = tile * V3_ROWS;
+            unsigned short* xb = x_base + (int)((tile - t_begin) % 3) * TBUF;
+            const bool more1 = tile + 1 < t_end;
+            const bool more2 = tile + 2 < t_end;
+
+            // [a] own tile's DMAs complete; newer: t+1's (8/RPD) if issued
+            if (more1) V3_ASM_VMCNT(8); else V3_ASM_VMCNT(0);
+            V3_BARRIER();
+            // y loads (hidden) BEFORE the prefetch, as in v3
+            const long long ymax = n_rows - 1;
+            long long yr0 = row0 + (threadIdx.x >> 4);
+            long long yr1 = row0 + 16 + (threadIdx.x >> 4);
+            const unsigned yb0 = v3_load_y_asm(y + (yr0 > ymax ? ymax : yr0));
+            const unsigned yb1 = v3_load_y_asm(y + (yr1 > ymax ? ymax : yr1));
+            if (more2) stage_tile((tile + 2) * V3_ROWS, (int)((tile + 2 - t_begin) % 3));
+
+            // ---- phase A: complete Z over this wave's K-quarter ----
+            f32x4_t z0 = {0.f, 0.f, 0.f, 0.f}, z1 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+            for (int ks = 0; ks < NKS; ++ks) {
+                const int kc = wid * QC + ks * 32;
+                frag_u bf, a0, a1;
+                const int g = (kc >> 3) + (lane >> 4);
+                const int arow0 = lane & 15;
+                const int arow1 = 16 + (lane & 15);
+                a0.q = *(U4*)&xb[arow0 * K + ((g ^ v3_key(arow0)) * 8)];
+                a1.q = *(U4*)&xb[arow1 * K + ((g ^ v3_key(arow1)) * 8)];
+                const int bk = kc + (lane >> 4) * 8;
+                bf.q = *(U4*)&th_lds[(lane & 15) * th_stride + bk];
+                z0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0.v, bf.v, z0, 0, 0, 0);
+                z1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1.v, bf.v, z1, 0, 0, 0);
+            }
+            {
+                float* zc = zc_lds + wid * V3_ROWS * BCH;
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    zc[((lane >> 4) * 4 + r) * BCH + (lane & 15)] = z0[r];
+                    zc[(16 + (lane >> 4) * 4 + r) * BCH + (lane & 15)] = z1[r];
+                }
+            }
+            // y complete: newer ops are the prefetch's DMAs only
+            if (more2) V3_ASM_VMCNT(8); else V3_ASM_VMCNT(0);
+            V3_BARRIER();
+            {
+#pragma unroll
+                for (int s = 0; s < 2; ++s) {
+                    const int slot = threadIdx.x + s * 256;
+                    const int row = slot >> 4;
+                    const int chain = slot & 15;
+                    float z = 0.f;
+#pragma unroll
+                    for (int w = 0; w < 4; ++w)
+                        z += zc_lds[(w * V3_ROWS + row) * BCH + chain];
+                    const long long grow = row0 + row;
+                    float resid = 0.f;
+                    if (grow < n_rows) {
+                        const float yv = bf16_bits_to_f32(
+                            (unsigned short)(s == 0 ? yb0 : yb1));
+                        const float sp = fmaxf(z, 0.f) + log1pf(__expf(-fabsf(z)));
+                        const float term = yv * z - sp;
+                        if (s == 0) logp0 += term; else logp1 += term;
+                        resid = yv - 1.f / (1.f + __expf(-z));
+                    }
+                    union { float f; unsigned int u; } cv;
+                    cv.f = resid;
+                    const unsigned int rnd = 0x7fff + ((cv.u >> 16) & 1);
+                    rt_lds[chain * rt_stride + row] =
+                        (unsigned short)((cv.u + rnd) >> 16);
+                }
+            }
+            V3_BARRIER();  // R visible
+
+            // ---- phase B over this wave's quarter ----
+#pragma unroll
+            for (int t2 = 0; t2 < NT2; ++t2) {
+                const int kc = wid * QC + t2 * 16;
+                const int kcol = kc + (lane & 15);
+                f32x4_t acc = g_acc[t2];
+                frag_u a, b;
+#pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    const int row = (lane >> 4) * 8 + j;
+                    a.u[j] = xb[row * K +
+                                (((kcol >> 3) ^ v3_key(row)) * 8) + (kcol & 7)];
+                }
+                b.q = *(U4*)&rt_lds[(lane & 15) * rt_stride + (lane >> 4) * 8];
+                g_acc[t2] =
+                    __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, acc, 0, 0, 0);
+            }
+            V3_BARRIER();  // xb free for the t+3 prefetch next iteration
+        }
+    }
+
+    // ---- epilogue ----
+    __syncthreads();
+    red_lds[threadIdx.x] = logp0 + logp1;
+    __syncthreads();
+    float* slab_blk = slab + (long long)blockIdx.x * (BCH + (long long)K * BCH);
+    if (threadIdx.x < BCH) {
+        float s = 0.f;
+        for (int i = threadIdx.x; i < 256; i += BCH) s += red_lds[i];
+        slab_blk[threadIdx.x] = s;
+    }
+    float* g_slab = slab_blk + BCH;
+#pragma unroll
+    for (int t2 = 0; t2 < NT2; ++t2) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int kcol = wid * QC + t2 * 16 + (lane >> 4) * 4 + r;
+            const int chain = lane & 15;
+            g_slab[(long long)kcol * BCH + chain] = g_acc[t2][r];
+        }
+    }
+}
+
 __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
     const unsigned short* __restrict__ X,   // [N][1024] bf16
     const unsigned short* __restrict__ y,   // [N] bf16
@@ -1868,16 +2061,23 @@ extern "C" int fed_logistic_glm_batched(
     // the v2 chunked kernel (which also serves K=512).
     const char* v3_env = getenv("FED_BATCHED_V3");
     const bool v3_on = v3_env ? atoi(v3_env) != 0 : true;
-    if (v3_on && !v1_variant && !lds_variant && K == 1024) {
+    if (v3_on && !v1_variant && !lds_variant && (K == 1024 || K == 512)) {
         // glds tile-resident variant: 1 block/CU, contiguous tile ranges
         int g3 = grid;
         if (g3 > 256) g3 = 256;
-        const int lds3 = (BCH * (K + TPAD) + 3 * V3_ROWS * V3_HALF +
+        const int xbytes = K == 1024 ? 3 * V3_ROWS * V3_HALF : 3 * V3_ROWS * 512;
+        const int lds3 = (BCH * (K + TPAD) + xbytes +
                           BCH * (V3_ROWS + RPAD)) * 2 +
                          (4 * V3_ROWS * BCH + 256) * 4 + 64;
-        hipLaunchKernelGGL(k_logistic_glm_batched_v3, dim3(g3), dim3(block), lds3,
-                           stream, (const unsigned short*)X, (const unsigned short*)y,
-                           n_rows, (const unsigned short*)theta_t_bf16, workspace);
+        if (K == 1024)
+            hipLaunchKernelGGL(k_logistic_glm_batched_v3, dim3(g3), dim3(block), lds3,
+                               stream, (const unsigned short*)X, (const unsigned short*)y,
+                               n_rows, (const unsigned short*)theta_t_bf16, workspace);
+        else
+            hipLaunchKernelGGL(k_logistic_glm_batched_v3s<512>, dim3(g3), dim3(block),
+                               lds3, stream, (const unsigned short*)X,
+                               (const unsigned short*)y, n_rows,
+                               (const unsigned short*)theta_t_bf16, workspace);
         hipError_t verr = hipGetLastError();
         if (verr != hipSuccess) return (int)verr;
         const int rg = ((int)slab_cols + 255) / 256;
