@@ -2057,10 +2057,12 @@ extern "C" int fed_logistic_glm_batched(
     const bool lds_variant = lds_env && atoi(lds_env) != 0;
     // v3 (glds tile-resident, 1x HBM traffic) is the default at K=1024:
     // 0.878 ms vs v2's 1.035 / v1's 1.568 at 2e6x1024x16, 5.17 ms at the
-    // config-4 shard (profiles/raw_r2/r2c10_*).  FED_BATCHED_V3=0 forces
-    // the v2 chunked kernel (which also serves K=512).
+    // config-4 shard (profiles/raw_r2/r2c10_*).  At K=512 the chunked v2
+    // stays default: its 64-KB tiles already re-read through L2 (measured
+    // 0.567 ms ~= the 1x floor vs v3s 0.645 at 2e6x512 -- raw_r2/r2c12);
+    // FED_BATCHED_V3=1/0 forces either way.
     const char* v3_env = getenv("FED_BATCHED_V3");
-    const bool v3_on = v3_env ? atoi(v3_env) != 0 : true;
+    const bool v3_on = v3_env ? atoi(v3_env) != 0 : (K == 1024);
     if (v3_on && !v1_variant && !lds_variant && (K == 1024 || K == 512)) {
         // glds tile-resident variant: 1 block/CU, contiguous tile ranges
         int g3 = grid;
