@@ -593,3 +593,28 @@ def test_mala_chains_over_native_ode(dev):
     assert 0.3 < stats["accept_rate"] < 0.9  # tuned toward 0.574
     post_mean = chain[100:].mean(axis=(0, 2))
     np.testing.assert_allclose(post_mean, theta_true, rtol=0.05)
+
+
+def test_batched_v3s_optin_matches_default_k512(dev):
+    """The opt-in K=512 tile-resident kernel (v3s) must track the default
+    chunked kernel's results (it is correct but not faster there; kept as
+    the FED_BATCHED_V3=1 path -- profiles/PROFILES.md K=512 note)."""
+    import os
+
+    X, y, _ = generate_logistic_dataset(300_000, 512, seed=91)
+    m = LogisticGLMModel(X, y, device=dev, dtype=torch.bfloat16)
+    theta = torch.randn(512, 16, device=dev,
+                        generator=torch.Generator(device=dev).manual_seed(6)) * 0.3
+    old = os.environ.pop("FED_BATCHED_V3", None)
+    try:
+        l2, g2 = m.logp_grad_batched(theta)
+        os.environ["FED_BATCHED_V3"] = "1"
+        l3, g3 = m.logp_grad_batched(theta)
+    finally:
+        os.environ.pop("FED_BATCHED_V3", None)
+        if old is not None:
+            os.environ["FED_BATCHED_V3"] = old
+    np.testing.assert_allclose(l3.cpu().numpy(), l2.cpu().numpy(), rtol=1e-6)
+    scale = float(g2.abs().max())
+    np.testing.assert_allclose(g3.cpu().numpy(), g2.cpu().numpy(),
+                               atol=1e-3 * scale, rtol=1e-3)
