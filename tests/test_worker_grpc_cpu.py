@@ -111,3 +111,30 @@ def test_fast_transport_echo_still_serves(echo_worker):
     outs = client.evaluate(np.float64(7.0), np.float64(-2.0))
     np.testing.assert_allclose(float(outs[0]), 5.0)
     del client
+
+
+@pytest.mark.timeout(120)
+def test_grpc_malformed_payload_is_clean_error(echo_worker):
+    """Garbage bytes in the gRPC message must come back as a clean INTERNAL
+    status (the worker's proto parser rejects them), not a crash or hang."""
+    import grpc
+
+    with grpc.insecure_channel(f"127.0.0.1:{PORT + 1}") as channel:
+        call = channel.unary_unary(
+            "/ArraysToArraysService/Evaluate",
+            request_serializer=lambda m: m,
+            response_deserializer=lambda b: b,
+        )
+        with pytest.raises(grpc.RpcError) as exc:
+            call(b"\xff\xfe\xfd garbage that is not a protobuf \x00\x01")
+        assert exc.value.code() in (
+            grpc.StatusCode.INTERNAL,
+            grpc.StatusCode.INVALID_ARGUMENT,
+        )
+    # the worker must still serve correct requests afterwards
+    from pytensor_federated_amd.service import ArraysToArraysServiceClient
+
+    client = ArraysToArraysServiceClient("127.0.0.1", PORT + 1, transport="grpc")
+    outs = client.evaluate(np.float64(1.0), np.float64(2.0))
+    np.testing.assert_allclose(float(outs[0]), 3.0)
+    del client
