@@ -2159,7 +2159,7 @@ extern "C" int fed_logistic_glm_batched(
 // Spin bounds. The REQUEST poll doubles as the idle lifetime: a resident
 // kernel blocks any device-wide synchronize, so it exits after ~1-2 s idle
 // and the host relaunches it transparently (hipStreamQuery detects exit).
-#define PK_REQ_SPIN_LIMIT 500000ll    // ~1-2 s idle -> self-exit
+#define PK_REQ_SPIN_LIMIT 2000000ll   // ~1-2 s idle at s_sleep(8) -> self-exit
 #define PK_SPIN_LIMIT 800000ll        // worker bcast poll (~1-2 s)
 #define PK_DONE_SPIN_LIMIT 200000ll   // completion barrier (compute is us-scale)
 
@@ -2187,7 +2187,7 @@ __global__ __launch_bounds__(256) void k_gaussian_persistent(
     double* __restrict__ slab,          // [grid][3] (ws)
     unsigned* __restrict__ ticket,      // sharded tickets (ws)
     PersistentState* __restrict__ st,   // device control block
-    const volatile double* __restrict__ req_host,  // pinned: [seq | a | b | quit]
+    const volatile double* __restrict__ req_host,  // pinned: [seq | a | b | -] (seq==SENTINEL -> quit)
     double* __restrict__ res_host       // pinned: [logp ga gb | seq]
 ) {
     using TR = VecTraits<T>;
@@ -2206,16 +2206,17 @@ __global__ __launch_bounds__(256) void k_gaussian_persistent(
         if (threadIdx.x == 0) {
             unsigned long long next = my_seq + 1;
             if (blockIdx.x == 0) {
-                // poll the HOST mailbox (one lane, one block)
+                // poll the HOST mailbox (one lane, one block).  Quit rides
+                // the seq word as PK_SENTINEL, so each poll is ONE host
+                // round trip; s_sleep(8) keeps the detection quantum ~0.2us
+                // (each poll read costs ~1us of host-memory latency anyway).
                 long long spins = 0;
                 while (true) {
                     const unsigned long long rs =
                         ((const volatile unsigned long long*)req_host)[0];
-                    const unsigned long long quit =
-                        ((const volatile unsigned long long*)req_host)[3];
-                    if (quit) { next = PK_SENTINEL; break; }
+                    if (rs == PK_SENTINEL) { next = PK_SENTINEL; break; }
                     if (rs >= next) break;
-                    __builtin_amdgcn_s_sleep(32);
+                    __builtin_amdgcn_s_sleep(8);
                     if (++spins > PK_REQ_SPIN_LIMIT) { next = PK_SENTINEL; break; }
                 }
                 if (next != PK_SENTINEL) {
@@ -2460,7 +2461,7 @@ int fed_gaussian_persistent_debug(void* handle, double* req8, double* res8) {
 
 int fed_gaussian_persistent_stop(void* handle) {
     FedPersistentLinear* e = (FedPersistentLinear*)handle;
-    __atomic_store_n((unsigned long long*)&e->req[3], 1ull, __ATOMIC_RELEASE);
+    __atomic_store_n((unsigned long long*)&e->req[0], PK_SENTINEL, __ATOMIC_RELEASE);
     hipError_t err = hipStreamSynchronize(e->stream);  // kernel exits on quit
     if (e->ws) (void)hipFree(e->ws);
     if (e->st) (void)hipFree(e->st);
