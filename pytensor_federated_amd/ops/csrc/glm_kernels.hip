@@ -2187,7 +2187,7 @@ __global__ __launch_bounds__(256) void k_gaussian_persistent(
     double* __restrict__ slab,          // [grid][3] (ws)
     unsigned* __restrict__ ticket,      // sharded tickets (ws)
     PersistentState* __restrict__ st,   // device control block
-    const volatile double* __restrict__ req_host,  // pinned: [seq | a | b | -] (seq==SENTINEL -> quit)
+    const volatile double* __restrict__ req_host,  // pinned seqlock: [seq | a | b | seq_pre] (seq==SENTINEL -> quit)
     double* __restrict__ res_host       // pinned: [logp ga gb | seq]
 ) {
     using TR = VecTraits<T>;
@@ -2206,24 +2206,38 @@ __global__ __launch_bounds__(256) void k_gaussian_persistent(
         if (threadIdx.x == 0) {
             unsigned long long next = my_seq + 1;
             if (blockIdx.x == 0) {
-                // poll the HOST mailbox (one lane, one block).  Quit rides
-                // the seq word as PK_SENTINEL, so each poll is ONE host
-                // round trip; s_sleep(8) keeps the detection quantum ~0.2us
-                // (each poll read costs ~1us of host-memory latency anyway).
+                // poll the HOST mailbox (one lane, one block).  SEQLOCK
+                // layout [seq | a | b | seq_pre]: the host writes seq_pre,
+                // then a, b, then seq (release), so a poll iteration whose
+                // four loads (issued together -- ONE host-memory round
+                // trip) sees seq >= next AND seq_pre == seq has a
+                // consistent payload in hand.  Quit rides seq as
+                // PK_SENTINEL.  This removes the separate post-detect
+                // payload read (~1us) and the second quit-flag read the
+                // round-1 protocol paid per poll.
                 long long spins = 0;
+                double a_req = 0.0, b_req = 0.0;
                 while (true) {
                     const unsigned long long rs =
                         ((const volatile unsigned long long*)req_host)[0];
+                    const double a_r = req_host[1];
+                    const double b_r = req_host[2];
+                    const unsigned long long pre =
+                        ((const volatile unsigned long long*)req_host)[3];
                     if (rs == PK_SENTINEL) { next = PK_SENTINEL; break; }
-                    if (rs >= next) break;
-                    __builtin_amdgcn_s_sleep(8);
+                    if (rs >= next && pre == rs) {
+                        a_req = a_r;
+                        b_req = b_r;
+                        break;
+                    }
+                    if (rs < next)  // no new request yet: back off
+                        __builtin_amdgcn_s_sleep(8);
+                    // torn read (host mid-write): immediate re-poll
                     if (++spins > PK_REQ_SPIN_LIMIT) { next = PK_SENTINEL; break; }
                 }
                 if (next != PK_SENTINEL) {
                     // sc1 payload + drained sc1 flag (G16 R1: a plain store
                     // + vmcnt drain is NOT cross-XCD visible)
-                    const double a_req = req_host[1];
-                    const double b_req = req_host[2];
                     store_sc1_f64(&st->theta[0], a_req);
                     store_sc1_f64(&st->theta[1], b_req);
                     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -2426,6 +2440,7 @@ void* fed_gaussian_persistent_start(
 int fed_gaussian_persistent_eval(void* handle, double a, double b, double* out3) {
     FedPersistentLinear* e = (FedPersistentLinear*)handle;
     e->seq += 1;
+    __atomic_store_n((unsigned long long*)&e->req[3], e->seq, __ATOMIC_RELEASE);
     e->req[1] = a;
     e->req[2] = b;
     __atomic_store_n((unsigned long long*)&e->req[0], e->seq, __ATOMIC_RELEASE);
