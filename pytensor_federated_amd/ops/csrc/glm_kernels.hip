@@ -2188,7 +2188,8 @@ __global__ __launch_bounds__(256) void k_gaussian_persistent(
     unsigned* __restrict__ ticket,      // sharded tickets (ws)
     PersistentState* __restrict__ st,   // device control block
     const volatile double* __restrict__ req_host,  // pinned seqlock: [seq | a | b | seq_pre] (seq==SENTINEL -> quit)
-    double* __restrict__ res_host       // pinned: [logp ga gb | seq]
+    double* __restrict__ res_host,      // pinned: [logp ga gb | seq]
+    int seqlock_on                      // 1: one-round-trip poll; 0: detect-then-read
 ) {
     using TR = VecTraits<T>;
     using A = typename TR::acc_t;
@@ -2217,23 +2218,38 @@ __global__ __launch_bounds__(256) void k_gaussian_persistent(
                 // round-1 protocol paid per poll.
                 long long spins = 0;
                 double a_req = 0.0, b_req = 0.0;
-                while (true) {
-                    const unsigned long long rs =
-                        ((const volatile unsigned long long*)req_host)[0];
-                    const double a_r = req_host[1];
-                    const double b_r = req_host[2];
-                    const unsigned long long pre =
-                        ((const volatile unsigned long long*)req_host)[3];
-                    if (rs == PK_SENTINEL) { next = PK_SENTINEL; break; }
-                    if (rs >= next && pre == rs) {
-                        a_req = a_r;
-                        b_req = b_r;
-                        break;
+                if (seqlock_on) {
+                    while (true) {
+                        const unsigned long long rs =
+                            ((const volatile unsigned long long*)req_host)[0];
+                        const double a_r = req_host[1];
+                        const double b_r = req_host[2];
+                        const unsigned long long pre =
+                            ((const volatile unsigned long long*)req_host)[3];
+                        if (rs == PK_SENTINEL) { next = PK_SENTINEL; break; }
+                        if (rs >= next && pre == rs) {
+                            a_req = a_r;
+                            b_req = b_r;
+                            break;
+                        }
+                        if (rs < next)  // no new request yet: back off
+                            __builtin_amdgcn_s_sleep(8);
+                        // torn read (host mid-write): immediate re-poll
+                        if (++spins > PK_REQ_SPIN_LIMIT) { next = PK_SENTINEL; break; }
                     }
-                    if (rs < next)  // no new request yet: back off
+                } else {
+                    while (true) {
+                        const unsigned long long rs =
+                            ((const volatile unsigned long long*)req_host)[0];
+                        if (rs == PK_SENTINEL) { next = PK_SENTINEL; break; }
+                        if (rs >= next) break;
                         __builtin_amdgcn_s_sleep(8);
-                    // torn read (host mid-write): immediate re-poll
-                    if (++spins > PK_REQ_SPIN_LIMIT) { next = PK_SENTINEL; break; }
+                        if (++spins > PK_REQ_SPIN_LIMIT) { next = PK_SENTINEL; break; }
+                    }
+                    if (next != PK_SENTINEL) {
+                        a_req = req_host[1];
+                        b_req = req_host[2];
+                    }
                 }
                 if (next != PK_SENTINEL) {
                     // sc1 payload + drained sc1 flag (G16 R1: a plain store
@@ -2362,24 +2378,30 @@ struct FedPersistentLinear {
 static int persistent_launch(FedPersistentLinear* e) {
     unsigned* ticket = (unsigned*)e->ws;
     double* slab = e->ws + 72;
+    // FED_PK_SEQLOCK=0 reverts to the detect-then-read poll for A/B
+    const char* sl = getenv("FED_PK_SEQLOCK");
+    const int seqlock_on = sl ? (atoi(sl) != 0) : 1;
     switch (e->dtype) {
         case FED_BF16:
             hipLaunchKernelGGL(k_gaussian_persistent<bf16_tag>, dim3(e->grid), dim3(256), 0,
                                e->stream, (const bf16_tag*)e->x, (const bf16_tag*)e->y, e->n,
                                e->inv_sig2, e->logp_const, slab, ticket, e->st,
-                               (const volatile double*)e->req_dev, (double*)e->res_dev);
+                               (const volatile double*)e->req_dev, (double*)e->res_dev,
+                               seqlock_on);
             break;
         case FED_F32:
             hipLaunchKernelGGL(k_gaussian_persistent<float>, dim3(e->grid), dim3(256), 0,
                                e->stream, (const float*)e->x, (const float*)e->y, e->n,
                                e->inv_sig2, e->logp_const, slab, ticket, e->st,
-                               (const volatile double*)e->req_dev, (double*)e->res_dev);
+                               (const volatile double*)e->req_dev, (double*)e->res_dev,
+                               seqlock_on);
             break;
         case FED_F64:
             hipLaunchKernelGGL(k_gaussian_persistent<double>, dim3(e->grid), dim3(256), 0,
                                e->stream, (const double*)e->x, (const double*)e->y, e->n,
                                e->inv_sig2, e->logp_const, slab, ticket, e->st,
-                               (const volatile double*)e->req_dev, (double*)e->res_dev);
+                               (const volatile double*)e->req_dev, (double*)e->res_dev,
+                               seqlock_on);
             break;
         default:
             return -2;
