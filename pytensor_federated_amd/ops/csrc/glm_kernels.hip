@@ -2189,7 +2189,8 @@ __global__ __launch_bounds__(256) void k_gaussian_persistent(
     PersistentState* __restrict__ st,   // device control block
     const volatile double* __restrict__ req_host,  // pinned seqlock: [seq | a | b | seq_pre] (seq==SENTINEL -> quit)
     double* __restrict__ res_host,      // pinned: [logp ga gb | seq]
-    int seqlock_on                      // 1: one-round-trip poll; 0: detect-then-read
+    int seqlock_on,                     // 1: one-round-trip poll; 0: detect-then-read
+    int fence_mode                      // 1: __threadfence_system publish; 0: store-drain
 ) {
     using TR = VecTraits<T>;
     using A = typename TR::acc_t;
@@ -2337,7 +2338,15 @@ __global__ __launch_bounds__(256) void k_gaussian_persistent(
                 rh[0] = logp_const - 0.5 * inv_sig2 * fin[0];
                 rh[1] = inv_sig2 * fin[1];
                 rh[2] = inv_sig2 * fin[2];
-                __threadfence_system();
+                if (fence_mode) {
+                    __threadfence_system();
+                } else {
+                    // posted writes to mapped host memory stay ordered on
+                    // the fabric; draining the store queue before the flag
+                    // store is sufficient and skips the full system fence
+                    // (A/B'd same-box; see profiles/PROFILES.md)
+                    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+                }
                 ((volatile unsigned long long*)res_host)[3] = my_seq;
                 store_sc1_u64(&st->done_seq, my_seq);
             }
@@ -2387,27 +2396,29 @@ static int persistent_launch(FedPersistentLinear* e) {
     // negative result.
     const char* sl = getenv("FED_PK_SEQLOCK");
     const int seqlock_on = sl ? (atoi(sl) != 0) : 0;
+    const char* fm = getenv("FED_PK_FENCE");
+    const int fence_mode = fm ? (atoi(fm) != 0) : 1;
     switch (e->dtype) {
         case FED_BF16:
             hipLaunchKernelGGL(k_gaussian_persistent<bf16_tag>, dim3(e->grid), dim3(256), 0,
                                e->stream, (const bf16_tag*)e->x, (const bf16_tag*)e->y, e->n,
                                e->inv_sig2, e->logp_const, slab, ticket, e->st,
                                (const volatile double*)e->req_dev, (double*)e->res_dev,
-                               seqlock_on);
+                               seqlock_on, fence_mode);
             break;
         case FED_F32:
             hipLaunchKernelGGL(k_gaussian_persistent<float>, dim3(e->grid), dim3(256), 0,
                                e->stream, (const float*)e->x, (const float*)e->y, e->n,
                                e->inv_sig2, e->logp_const, slab, ticket, e->st,
                                (const volatile double*)e->req_dev, (double*)e->res_dev,
-                               seqlock_on);
+                               seqlock_on, fence_mode);
             break;
         case FED_F64:
             hipLaunchKernelGGL(k_gaussian_persistent<double>, dim3(e->grid), dim3(256), 0,
                                e->stream, (const double*)e->x, (const double*)e->y, e->n,
                                e->inv_sig2, e->logp_const, slab, ticket, e->st,
                                (const volatile double*)e->req_dev, (double*)e->res_dev,
-                               seqlock_on);
+                               seqlock_on, fence_mode);
             break;
         default:
             return -2;
