@@ -2341,10 +2341,12 @@ __global__ __launch_bounds__(256) void k_gaussian_persistent(
                 if (fence_mode) {
                     __threadfence_system();
                 } else {
-                    // posted writes to mapped host memory stay ordered on
-                    // the fabric; draining the store queue before the flag
-                    // store is sufficient and skips the full system fence
-                    // (A/B'd same-box; see profiles/PROFILES.md)
+                    // store-drain instead of the system fence: same-box A/B
+                    // measured 55.4-55.8k vs 54.7-55.7k calls/s -- ~1%,
+                    // within rep noise, so the formally correct
+                    // __threadfence_system stays the default and this
+                    // path is kept only as the documented experiment
+                    // (FED_PK_FENCE=0)
                     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
                 }
                 ((volatile unsigned long long*)res_host)[3] = my_seq;
