@@ -121,7 +121,7 @@ class TestPolyODEModelCPU:
                 logp_ref = logp_ref - (r * r).sum() / (2 * sig2)
         logp_ref = logp_ref - 0.5 * yt.numel() * np.log(2 * np.pi * sig2)
         (g_ref,) = torch.autograd.grad(logp_ref, th)
-        np.testing.assert_allclose(float(logp), float(logp_ref), rtol=1e-12)
+        np.testing.assert_allclose(float(logp), float(logp_ref.detach()), rtol=1e-12)
         np.testing.assert_allclose(g.numpy(), g_ref.numpy(), rtol=1e-10)
 
     def test_batched_eager_loops_chains(self):
@@ -204,7 +204,7 @@ class TestRandomTableProperty:
                 g_ref = None
             if g_ref is None:
                 g_ref = torch.zeros_like(th)
-            np.testing.assert_allclose(float(logp), float(logp_ref), rtol=1e-11,
+            np.testing.assert_allclose(float(logp), float(logp_ref.detach()), rtol=1e-11,
                                        err_msg=f"trial {trial} terms={terms}")
             np.testing.assert_allclose(
                 g.numpy(), g_ref.numpy(), rtol=1e-8, atol=1e-10,
