@@ -1405,14 +1405,24 @@ __device__ __forceinline__ int v3_key(int row) {
 // leaves its completion entirely to this kernel's hand-counted
 // V3_ASM_VMCNT waits (the glds queue shares vmcnt, FIFO).
 __device__ __forceinline__ void v3_glds_row(const void* src, unsigned lds_byte_off) {
+    // nt: each X line is streamed by exactly one CU exactly once (the
+    // tile-resident design's whole point), the guide's nt-weights case
+    // (issued->landed -18%).  No "memory" clobber: it would make hipcc
+    // bundle a conservative vmcnt(0) into every later ds_read's wait (the
+    // exact drain this asm form exists to avoid) -- ordering is carried
+    // by the volatile asm barriers (which DO clobber memory).
+    asm volatile(
+        "s_mov_b32 m0, %0\n\t"
+        "s_nop 0\n\t"
+        "global_load_lds_dwordx4 %1, off nt"
+        :: "s"(lds_byte_off), "v"(src));
+}
+
+__device__ __forceinline__ void v3_glds_row_cached(const void* src, unsigned lds_byte_off) {
     asm volatile(
         "s_mov_b32 m0, %0\n\t"
         "s_nop 0\n\t"
         "global_load_lds_dwordx4 %1, off"
-        // no "memory" clobber: it would make hipcc bundle a conservative
-        // vmcnt(0) into every later ds_read's wait (the exact drain this
-        // asm form exists to avoid).  Ordering is carried by the volatile
-        // asm barriers (which DO clobber memory) on either side.
         :: "s"(lds_byte_off), "v"(src));
 }
 
@@ -1434,7 +1444,7 @@ __device__ __forceinline__ unsigned v3_load_y_asm(const unsigned short* addr) {
 __device__ __forceinline__ void v3_stage_half(
     const unsigned short* __restrict__ X, long long n_rows, int K,
     long long row0, int colbase, unsigned short* half_buf,
-    const char* smem_base, int wid, int lane
+    const char* smem_base, int wid, int lane, int nt_on
 ) {
 #pragma unroll
     for (int rr = 0; rr < 8; ++rr) {
@@ -1446,8 +1456,10 @@ __device__ __forceinline__ void v3_stage_half(
                 &X[grow * (long long)K + colbase + ((lane ^ v3_key(r)) * 8)];
             // the row base is wave-uniform by construction; readfirstlane
             // makes that provable so the asm "s" constraint gets an SGPR
-            v3_glds_row(src, __builtin_amdgcn_readfirstlane(
-                                 (unsigned)((const char*)dst - smem_base)));
+            const unsigned off = __builtin_amdgcn_readfirstlane(
+                (unsigned)((const char*)dst - smem_base));
+            if (nt_on) v3_glds_row(src, off);
+            else v3_glds_row_cached(src, off);
         } else {
             *(U4*)&dst[lane * 8] = (U4){0, 0, 0, 0};
         }
@@ -1466,7 +1478,8 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3s(
     const unsigned short* __restrict__ y,   // [N] bf16
     long long n_rows,
     const unsigned short* __restrict__ theta_t,  // [16][K] bf16
-    float* __restrict__ slab                     // [grid][16 + K*16]
+    float* __restrict__ slab,                    // [grid][16 + K*16]
+    int nt_on                                    // nt on the tile DMAs (A/B)
 ) {
     const int lane = threadIdx.x & 63;
     const int wid = threadIdx.x >> 6;
@@ -1518,8 +1531,10 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3s(
             if (grow < n_rows) {
                 const unsigned short* src =
                     &X[grow * (long long)K + ((lane ^ v3_key(r)) * 8)];
-                v3_glds_row(src, __builtin_amdgcn_readfirstlane(
-                                     (unsigned)((const char*)dst - smem)));
+                const unsigned off3 = __builtin_amdgcn_readfirstlane(
+                    (unsigned)((const char*)dst - smem));
+                if (nt_on) v3_glds_row(src, off3);
+                else v3_glds_row_cached(src, off3);
             } else {
                 *(U4*)&dst[lane * 8] = (U4){0, 0, 0, 0};
             }
@@ -1652,7 +1667,8 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
     const unsigned short* __restrict__ y,   // [N] bf16
     long long n_rows,
     const unsigned short* __restrict__ theta_t,  // [16][1024] bf16
-    float* __restrict__ slab                     // [grid][16 + 1024*16]
+    float* __restrict__ slab,                    // [grid][16 + 1024*16]
+    int nt_on                                    // nt on the tile DMAs (A/B)
 ) {
     constexpr int K = 2 * V3_HALF;
     const int lane = threadIdx.x & 63;
@@ -1698,8 +1714,8 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
     if (t_begin < t_end) {
         // prologue: stage tile t_begin fully.  h0 buffers alternate 0/2,
         // h1 lives in buffer 1.
-        v3_stage_half(X, n_rows, K, t_begin * V3_ROWS, 0, x_base, smem, wid, lane);
-        v3_stage_half(X, n_rows, K, t_begin * V3_ROWS, V3_HALF, x_base + HBUF, smem, wid, lane);
+        v3_stage_half(X, n_rows, K, t_begin * V3_ROWS, 0, x_base, smem, wid, lane, nt_on);
+        v3_stage_half(X, n_rows, K, t_begin * V3_ROWS, V3_HALF, x_base + HBUF, smem, wid, lane, nt_on);
 
         int h0sel = 0;  // buffer index (0 or 2) holding the CURRENT tile's h0
 #pragma unroll 1
@@ -1724,7 +1740,7 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
             const unsigned yb1 = v3_load_y_asm(y + (yr1 > ymax ? ymax : yr1));
             // [2] prefetch next tile's h0 as deep as possible
             if (more)
-                v3_stage_half(X, n_rows, K, row0 + V3_ROWS, 0, h0n, smem, wid, lane);
+                v3_stage_half(X, n_rows, K, row0 + V3_ROWS, 0, h0n, smem, wid, lane, nt_on);
 
             // ---- phase A on h0: z += X[:, w*128 .. +128) . theta ----
             f32x4_t z0 = {0.f, 0.f, 0.f, 0.f}, z1 = {0.f, 0.f, 0.f, 0.f};
@@ -1823,7 +1839,7 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
             V3_BARRIER();  // everyone done reading h1
             // [6] refill h1 with the NEXT tile's second half
             if (more)
-                v3_stage_half(X, n_rows, K, row0 + V3_ROWS, V3_HALF, h1, smem, wid, lane);
+                v3_stage_half(X, n_rows, K, row0 + V3_ROWS, V3_HALF, h1, smem, wid, lane, nt_on);
 
             // ---- phase B, h0 columns ----
 #pragma unroll
@@ -2071,15 +2087,18 @@ extern "C" int fed_logistic_glm_batched(
         const int lds3 = (BCH * (K + TPAD) + xbytes +
                           BCH * (V3_ROWS + RPAD)) * 2 +
                          (4 * V3_ROWS * BCH + 256) * 4 + 64;
+        const char* nt = getenv("FED_V3_NT");
+        const int nt_on = nt ? (atoi(nt) != 0) : 1;  // stream-once data: nt default
         if (K == 1024)
             hipLaunchKernelGGL(k_logistic_glm_batched_v3, dim3(g3), dim3(block), lds3,
                                stream, (const unsigned short*)X, (const unsigned short*)y,
-                               n_rows, (const unsigned short*)theta_t_bf16, workspace);
+                               n_rows, (const unsigned short*)theta_t_bf16, workspace,
+                               nt_on);
         else
             hipLaunchKernelGGL(k_logistic_glm_batched_v3s<512>, dim3(g3), dim3(block),
                                lds3, stream, (const unsigned short*)X,
                                (const unsigned short*)y, n_rows,
-                               (const unsigned short*)theta_t_bf16, workspace);
+                               (const unsigned short*)theta_t_bf16, workspace, nt_on);
         hipError_t verr = hipGetLastError();
         if (verr != hipSuccess) return (int)verr;
         const int rg = ((int)slab_cols + 255) / 256;
