@@ -68,4 +68,4 @@ try:
 except ModuleNotFoundError:
     pass
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"

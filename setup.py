@@ -28,7 +28,7 @@ class BuildHip(Command):
 
 setup(
     name="pytensor-federated-amd",
-    version="0.1.0",
+    version="0.2.0",
     description="MI355X-native federated logp/gradient engine",
     packages=find_packages(include=["pytensor_federated_amd*"]),
     package_data={"pytensor_federated_amd.ops": ["*.so", "csrc/*.hip"]},
