@@ -48,7 +48,7 @@ try:
         if key not in refs:
             refs[key] = ref(a, b)
         for _ in range(5):
-            logp, (ga, gb) = c.evaluate(a, b)
+            logp, ga, gb = c.evaluate(a, b)
             lr, (gar, gbr) = refs[key]
             np.testing.assert_allclose(float(logp), float(lr), rtol=1e-9)
             np.testing.assert_allclose(float(ga), float(gar), rtol=1e-7)
