@@ -1458,7 +1458,7 @@ __device__ __forceinline__ void v3_stage_half(
             // makes that provable so the asm "s" constraint gets an SGPR
             const unsigned off = __builtin_amdgcn_readfirstlane(
                 (unsigned)((const char*)dst - smem_base));
-            if (nt_on) v3_glds_row(src, off);
+            if (nt_on & 1) v3_glds_row(src, off);
             else v3_glds_row_cached(src, off);
         } else {
             *(U4*)&dst[lane * 8] = (U4){0, 0, 0, 0};
@@ -1533,7 +1533,7 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3s(
                     &X[grow * (long long)K + ((lane ^ v3_key(r)) * 8)];
                 const unsigned off3 = __builtin_amdgcn_readfirstlane(
                     (unsigned)((const char*)dst - smem));
-                if (nt_on) v3_glds_row(src, off3);
+                if (nt_on & 1) v3_glds_row(src, off3);
                 else v3_glds_row_cached(src, off3);
             } else {
                 *(U4*)&dst[lane * 8] = (U4){0, 0, 0, 0};
@@ -1668,11 +1668,19 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
     long long n_rows,
     const unsigned short* __restrict__ theta_t,  // [16][1024] bf16
     float* __restrict__ slab,                    // [grid][16 + 1024*16]
-    int nt_on                                    // nt on the tile DMAs (A/B)
+    int nt_on                                    // bit0: nt DMAs; bit1: profile
 ) {
     constexpr int K = 2 * V3_HALF;
     const int lane = threadIdx.x & 63;
     const int wid = threadIdx.x >> 6;
+    // FED_V3_PROF: wid-0 lane-0 of every block accumulates cycle counts
+    // for the two stage-gate waits, the mid (zc/R) barrier region and the
+    // phase-B h1 barrier into slab rows past the v3 grid (256 blocks x 8
+    // u64 in otherwise-unused workspace) -- splits PMC's "35% wait" into
+    // stage-starvation vs barrier-skew without touching the default path.
+    const int prof_on = (nt_on >> 1) & 1;
+    unsigned long long p_gate1 = 0, p_gate4 = 0, p_mid = 0, p_bh = 0, p_t0 = 0;
+    if (prof_on && wid == 0 && lane == 0) p_t0 = __builtin_amdgcn_s_memtime();
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
     unsigned short* th_lds = (unsigned short*)smem;        // [16][K+TPAD]
@@ -1727,8 +1735,12 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
             const bool more = tile + 1 < t_end;
 
             // [1] own h0 DMAs done (h1's 8 may stay in flight), all waves
+            unsigned long long pt = 0;
+            if (prof_on && wid == 0 && lane == 0) pt = __builtin_amdgcn_s_memtime();
             V3_ASM_VMCNT(8);
             V3_BARRIER();
+            if (prof_on && wid == 0 && lane == 0)
+                p_gate1 += __builtin_amdgcn_s_memtime() - pt;
             // this thread's two y values (R-step slots), hidden loads on
             // the same queue, issued BEFORE the h0 prefetch so the [4]
             // counted wait (which leaves only the prefetch outstanding)
@@ -1763,8 +1775,11 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
             // was actually issued after them: 8 prefetch DMAs on interior
             // tiles, NOTHING on the last tile (an unconditional vmcnt(8)
             // there would leave h1/y unwaited -- a timing-dependent race).
+            if (prof_on && wid == 0 && lane == 0) pt = __builtin_amdgcn_s_memtime();
             if (more) V3_ASM_VMCNT(8); else V3_ASM_VMCNT(0);
             V3_BARRIER();
+            if (prof_on && wid == 0 && lane == 0)
+                p_gate4 += __builtin_amdgcn_s_memtime() - pt;
             // ---- phase A on h1 ----
 #pragma unroll
             for (int ks = 0; ks < 4; ++ks) {
@@ -1789,7 +1804,10 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
                     zc[(16 + (lane >> 4) * 4 + r) * BCH + (lane & 15)] = z1[r];
                 }
             }
+            if (prof_on && wid == 0 && lane == 0) pt = __builtin_amdgcn_s_memtime();
             V3_BARRIER();
+            if (prof_on && wid == 0 && lane == 0)
+                p_mid += __builtin_amdgcn_s_memtime() - pt;
             {
 #pragma unroll
                 for (int s = 0; s < 2; ++s) {
@@ -1817,7 +1835,10 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
                         (unsigned short)((cv.u + rnd) >> 16);
                 }
             }
+            if (prof_on && wid == 0 && lane == 0) pt = __builtin_amdgcn_s_memtime();
             V3_BARRIER();  // R visible
+            if (prof_on && wid == 0 && lane == 0)
+                p_mid += __builtin_amdgcn_s_memtime() - pt;
 
             // ---- phase B, h1 columns first (frees h1 for the refill) ----
 #pragma unroll
@@ -1836,7 +1857,10 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
                 g_acc[8 + t2] =
                     __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, acc, 0, 0, 0);
             }
+            if (prof_on && wid == 0 && lane == 0) pt = __builtin_amdgcn_s_memtime();
             V3_BARRIER();  // everyone done reading h1
+            if (prof_on && wid == 0 && lane == 0)
+                p_bh += __builtin_amdgcn_s_memtime() - pt;
             // [6] refill h1 with the NEXT tile's second half
             if (more)
                 v3_stage_half(X, n_rows, K, row0 + V3_ROWS, V3_HALF, h1, smem, wid, lane, nt_on);
@@ -1862,6 +1886,17 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
         }
     }
 
+    if (prof_on && wid == 0 && lane == 0) {
+        // rows >= 300 of the workspace slab are unused at v3's grid (256)
+        unsigned long long* prof =
+            (unsigned long long*)(slab + 300LL * (BCH + (long long)K * BCH)) +
+            8LL * blockIdx.x;
+        prof[0] = p_gate1;
+        prof[1] = p_gate4;
+        prof[2] = p_mid;
+        prof[3] = p_bh;
+        prof[4] = __builtin_amdgcn_s_memtime() - p_t0;
+    }
     // ---- epilogue: block partials -> slab (layout shared with v1/v2) ----
     __syncthreads();
     red_lds[threadIdx.x] = logp0 + logp1;
@@ -2088,7 +2123,9 @@ extern "C" int fed_logistic_glm_batched(
                           BCH * (V3_ROWS + RPAD)) * 2 +
                          (4 * V3_ROWS * BCH + 256) * 4 + 64;
         const char* nt = getenv("FED_V3_NT");
-        const int nt_on = nt ? (atoi(nt) != 0) : 1;  // stream-once data: nt default
+        int nt_on = nt ? (atoi(nt) != 0) : 1;  // stream-once data: nt default
+        const char* pf = getenv("FED_V3_PROF");
+        if (pf && atoi(pf) != 0) nt_on |= 2;
         if (K == 1024)
             hipLaunchKernelGGL(k_logistic_glm_batched_v3, dim3(g3), dim3(block), lds3,
                                stream, (const unsigned short*)X, (const unsigned short*)y,
