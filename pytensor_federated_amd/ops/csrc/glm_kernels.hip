@@ -1680,6 +1680,7 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
     // stage-starvation vs barrier-skew without touching the default path.
     const int prof_on = (nt_on >> 1) & 1;
     unsigned long long p_gate1 = 0, p_gate4 = 0, p_mid = 0, p_bh = 0, p_t0 = 0;
+    unsigned long long p_phA = 0, p_phB = 0;
     if (prof_on && wid == 0 && lane == 0) p_t0 = __builtin_amdgcn_s_memtime();
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -1755,6 +1756,8 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
                 v3_stage_half(X, n_rows, K, row0 + V3_ROWS, 0, h0n, smem, wid, lane, nt_on);
 
             // ---- phase A on h0: z += X[:, w*128 .. +128) . theta ----
+            unsigned long long pa = 0;
+            if (prof_on && wid == 0 && lane == 0) pa = __builtin_amdgcn_s_memtime();
             f32x4_t z0 = {0.f, 0.f, 0.f, 0.f}, z1 = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
             for (int ks = 0; ks < 4; ++ks) {
@@ -1771,6 +1774,8 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
                 b1f.q = b0f.q;
                 z1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1.v, b1f.v, z1, 0, 0, 0);
             }
+            if (prof_on && wid == 0 && lane == 0)
+                p_phA += __builtin_amdgcn_s_memtime() - pa;
             // [4] own h1 DMAs + y loads done.  The count must match what
             // was actually issued after them: 8 prefetch DMAs on interior
             // tiles, NOTHING on the last tile (an unconditional vmcnt(8)
@@ -1780,6 +1785,7 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
             V3_BARRIER();
             if (prof_on && wid == 0 && lane == 0)
                 p_gate4 += __builtin_amdgcn_s_memtime() - pt;
+            if (prof_on && wid == 0 && lane == 0) pa = __builtin_amdgcn_s_memtime();
             // ---- phase A on h1 ----
 #pragma unroll
             for (int ks = 0; ks < 4; ++ks) {
@@ -1795,6 +1801,8 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
                 z0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0.v, bf.v, z0, 0, 0, 0);
                 z1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1.v, bf.v, z1, 0, 0, 0);
             }
+            if (prof_on && wid == 0 && lane == 0)
+                p_phA += __builtin_amdgcn_s_memtime() - pa;
             // ---- combine the 4 waves' z quarters; logp + R ----
             {
                 float* zc = zc_lds + wid * V3_ROWS * BCH;
@@ -1840,6 +1848,7 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
             if (prof_on && wid == 0 && lane == 0)
                 p_mid += __builtin_amdgcn_s_memtime() - pt;
 
+            if (prof_on && wid == 0 && lane == 0) pa = __builtin_amdgcn_s_memtime();
             // ---- phase B, h1 columns first (frees h1 for the refill) ----
 #pragma unroll
             for (int t2 = 0; t2 < 8; ++t2) {
@@ -1857,7 +1866,10 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
                 g_acc[8 + t2] =
                     __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, acc, 0, 0, 0);
             }
-            if (prof_on && wid == 0 && lane == 0) pt = __builtin_amdgcn_s_memtime();
+            if (prof_on && wid == 0 && lane == 0) {
+                p_phB += __builtin_amdgcn_s_memtime() - pa;
+                pt = __builtin_amdgcn_s_memtime();
+            }
             V3_BARRIER();  // everyone done reading h1
             if (prof_on && wid == 0 && lane == 0)
                 p_bh += __builtin_amdgcn_s_memtime() - pt;
@@ -1865,6 +1877,7 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
             if (more)
                 v3_stage_half(X, n_rows, K, row0 + V3_ROWS, V3_HALF, h1, smem, wid, lane, nt_on);
 
+            if (prof_on && wid == 0 && lane == 0) pa = __builtin_amdgcn_s_memtime();
             // ---- phase B, h0 columns ----
 #pragma unroll
             for (int t2 = 0; t2 < 8; ++t2) {
@@ -1882,6 +1895,8 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
                 g_acc[t2] =
                     __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, acc, 0, 0, 0);
             }
+            if (prof_on && wid == 0 && lane == 0)
+                p_phB += __builtin_amdgcn_s_memtime() - pa;
             h0sel ^= 2;
         }
     }
@@ -1896,6 +1911,8 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
         prof[2] = p_mid;
         prof[3] = p_bh;
         prof[4] = __builtin_amdgcn_s_memtime() - p_t0;
+        prof[5] = p_phA;
+        prof[6] = p_phB;
     }
     // ---- epilogue: block partials -> slab (layout shared with v1/v2) ----
     __syncthreads();

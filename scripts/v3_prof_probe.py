@@ -39,6 +39,10 @@ res = {
     "gate4_h1y_pct": float((p[:, 1] / p[:, 4]).mean() * 100),
     "zcR_barriers_pct": float((p[:, 2] / p[:, 4]).mean() * 100),
     "phaseB_h1_barrier_pct": float((p[:, 3] / p[:, 4]).mean() * 100),
+    "phaseA_pct": float((p[:, 5] / p[:, 4]).mean() * 100),
+    "phaseB_pct": float((p[:, 6] / p[:, 4]).mean() * 100),
 }
-res["accounted_wait_pct"] = sum(v for k, v in res.items() if k.endswith("_pct"))
+res["accounted_wait_pct"] = sum(
+    res[k] for k in ("gate1_h0_pct", "gate4_h1y_pct", "zcR_barriers_pct",
+                     "phaseB_h1_barrier_pct"))
 print(json.dumps(res))
