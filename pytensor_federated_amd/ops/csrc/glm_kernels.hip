@@ -1662,6 +1662,7 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3s(
     }
 }
 
+template <int PROF>
 __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
     const unsigned short* __restrict__ X,   // [N][1024] bf16
     const unsigned short* __restrict__ y,   // [N] bf16
@@ -1678,7 +1679,8 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
     // phase-B h1 barrier into slab rows past the v3 grid (256 blocks x 8
     // u64 in otherwise-unused workspace) -- splits PMC's "35% wait" into
     // stage-starvation vs barrier-skew without touching the default path.
-    const int prof_on = (nt_on >> 1) & 1;
+    constexpr int prof_on = PROF;  // compile-time: the default instantiation
+                                   // carries ZERO instrumentation branches
     unsigned long long p_gate1 = 0, p_gate4 = 0, p_mid = 0, p_bh = 0, p_t0 = 0;
     unsigned long long p_phA = 0, p_phB = 0;
     if (prof_on && wid == 0 && lane == 0) p_t0 = __builtin_amdgcn_s_memtime();
@@ -2143,11 +2145,18 @@ extern "C" int fed_logistic_glm_batched(
         int nt_on = nt ? (atoi(nt) != 0) : 1;  // stream-once data: nt default
         const char* pf = getenv("FED_V3_PROF");
         if (pf && atoi(pf) != 0) nt_on |= 2;
-        if (K == 1024)
-            hipLaunchKernelGGL(k_logistic_glm_batched_v3, dim3(g3), dim3(block), lds3,
-                               stream, (const unsigned short*)X, (const unsigned short*)y,
-                               n_rows, (const unsigned short*)theta_t_bf16, workspace,
-                               nt_on);
+        if (K == 1024) {
+            if (nt_on & 2)
+                hipLaunchKernelGGL(k_logistic_glm_batched_v3<1>, dim3(g3), dim3(block),
+                                   lds3, stream, (const unsigned short*)X,
+                                   (const unsigned short*)y, n_rows,
+                                   (const unsigned short*)theta_t_bf16, workspace, nt_on);
+            else
+                hipLaunchKernelGGL(k_logistic_glm_batched_v3<0>, dim3(g3), dim3(block),
+                                   lds3, stream, (const unsigned short*)X,
+                                   (const unsigned short*)y, n_rows,
+                                   (const unsigned short*)theta_t_bf16, workspace, nt_on);
+        }
         else
             hipLaunchKernelGGL(k_logistic_glm_batched_v3s<512>, dim3(g3), dim3(block),
                                lds3, stream, (const unsigned short*)X,
