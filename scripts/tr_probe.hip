@@ -25,10 +25,11 @@ __global__ void k_tr_probe(unsigned short* out, int pattern) {
         case 2: eoff = (lane >> 4) * 64 + (lane & 15) * 4; break;
         default: eoff = lane * 4; break;
     }
-    const unsigned short* addr = &lds[eoff];
+    // DS vaddr is a 32-bit LDS byte address (static __shared__ starts at 0)
+    const unsigned a32 = (unsigned)((const char*)&lds[eoff] - (const char*)&lds[0]);
     unsigned long long v;
     asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
-                 : "=v"(v) : "v"(addr));
+                 : "=v"(v) : "v"(a32));
     if (threadIdx.x < 64) {
         out[lane * 4 + 0] = (unsigned short)(v & 0xffff);
         out[lane * 4 + 1] = (unsigned short)((v >> 16) & 0xffff);
