@@ -23,13 +23,20 @@ __global__ void k_tr_probe(unsigned short* out, int pattern) {
         case 0: eoff = lane * 4; break;                       // 8-B pieces, linear
         case 1: eoff = (lane & 15) + (lane >> 4) * 64; break; // canonical formula
         case 2: eoff = (lane >> 4) * 64 + (lane & 15) * 4; break;
+        case 3: eoff = lane * 4; break;  // control via plain ds_read_b64
         default: eoff = lane * 4; break;
     }
     // DS vaddr is a 32-bit LDS byte address (static __shared__ starts at 0)
     const unsigned a32 = (unsigned)((const char*)&lds[eoff] - (const char*)&lds[0]);
     unsigned long long v;
-    asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
-                 : "=v"(v) : "v"(a32));
+    if (pattern == 3) {
+        // control: plain b64 through the identical asm path
+        asm volatile("ds_read_b64 %0, %1\n\ts_waitcnt lgkmcnt(0) vmcnt(0)"
+                     : "=v"(v) : "v"(a32));
+    } else {
+        asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0) vmcnt(0)"
+                     : "=v"(v) : "v"(a32));
+    }
     if (threadIdx.x < 64) {
         out[lane * 4 + 0] = (unsigned short)(v & 0xffff);
         out[lane * 4 + 1] = (unsigned short)((v >> 16) & 0xffff);
@@ -42,7 +49,7 @@ int main() {
     unsigned short* out_d;
     (void)hipMalloc(&out_d, 64 * 4 * 2);
     unsigned short out_h[256];
-    for (int pat = 0; pat < 3; ++pat) {
+    for (int pat = 0; pat < 4; ++pat) {
         hipLaunchKernelGGL(k_tr_probe, dim3(1), dim3(64), 0, 0, out_d, pat);
         if (hipDeviceSynchronize() != hipSuccess) { printf("pat %d: launch failed\n", pat); continue; }
         (void)hipMemcpy(out_h, out_d, sizeof(out_h), hipMemcpyDeviceToHost);
