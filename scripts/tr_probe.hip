@@ -32,10 +32,10 @@ __global__ void k_tr_probe(unsigned short* out, int pattern) {
     if (pattern == 3) {
         // control: plain b64 through the identical asm path
         asm volatile("ds_read_b64 %0, %1\n\ts_waitcnt lgkmcnt(0) vmcnt(0)"
-                     : "=v"(v) : "v"(a32));
+                     : "=v"(v) : "v"(a32) : "memory");
     } else {
         asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0) vmcnt(0)"
-                     : "=v"(v) : "v"(a32));
+                     : "=v"(v) : "v"(a32) : "memory");
     }
     if (threadIdx.x < 64) {
         out[lane * 4 + 0] = (unsigned short)(v & 0xffff);
