@@ -37,6 +37,9 @@ __global__ void k_tr_probe(unsigned short* out, int pattern) {
         asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0) vmcnt(0)"
                      : "=v"(v) : "v"(a32) : "memory");
     }
+    // a plain C read of the array keeps the fill alive (the asm only sees
+    // an integer address, so LLVM otherwise proves the stores dead)
+    out[256 + threadIdx.x] = lds[threadIdx.x];
     if (threadIdx.x < 64) {
         out[lane * 4 + 0] = (unsigned short)(v & 0xffff);
         out[lane * 4 + 1] = (unsigned short)((v >> 16) & 0xffff);
@@ -47,7 +50,7 @@ __global__ void k_tr_probe(unsigned short* out, int pattern) {
 
 int main() {
     unsigned short* out_d;
-    (void)hipMalloc(&out_d, 64 * 4 * 2);
+    (void)hipMalloc(&out_d, (256 + 64) * 2 * 2);
     unsigned short out_h[256];
     for (int pat = 0; pat < 4; ++pat) {
         hipLaunchKernelGGL(k_tr_probe, dim3(1), dim3(64), 0, 0, out_d, pat);
