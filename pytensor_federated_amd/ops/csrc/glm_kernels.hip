@@ -1466,6 +1466,45 @@ __device__ __forceinline__ void v3_stage_half(
     }
 }
 
+// v4 staging: one 1-KB DMA = 4 rows x 128 cols landing as 8 contiguous
+// [4][16] subtiles; per wave 8 DMAs (row-groups 2w..2w+1 x 4 col-octets),
+// so the v3 vmcnt ledgers hold unchanged.  Lane n supplies the source of
+// image element n*8: subtile s = n>>3, row j = (n&7)>>1, col-half h = n&1.
+__device__ __forceinline__ void v4_stage_half(
+    const unsigned short* __restrict__ X, long long n_rows, int K,
+    long long row0, int colbase, unsigned short* half_buf,
+    const char* smem_base, int wid, int lane, int nt_on
+) {
+#pragma unroll
+    for (int rr = 0; rr < 8; ++rr) {
+        const int rg = wid * 2 + (rr >> 2);   // row-group (4 rows) 0..7
+        const int oct = rr & 3;               // col-octet (128 cols) 0..3
+        const long long grow0 = row0 + rg * 4;
+        unsigned short* dst = half_buf + (rg * 32 + oct * 8) * 64;
+        const int s = lane >> 3;
+        const int j = (lane & 7) >> 1;
+        const int h = lane & 1;
+        if (grow0 + 4 <= n_rows) {
+            const unsigned short* srcp =
+                &X[(grow0 + j) * (long long)K + colbase + oct * 128 + s * 16 + h * 8];
+            const unsigned off = __builtin_amdgcn_readfirstlane(
+                (unsigned)((const char*)dst - smem_base));
+            if (nt_on & 1) v3_glds_row(srcp, off);
+            else v3_glds_row_cached(srcp, off);
+        } else {
+            // edge row-group: zero-fill, then guarded plain writes of the
+            // in-range rows through the subtiled mapping
+            *(U4*)&dst[lane * 8] = (U4){0, 0, 0, 0};
+            const long long grow = grow0 + j;
+            if (grow < n_rows) {
+                *(U4*)&dst[(s * 64) + (j * 16) + h * 8] =
+                    *(const U4*)&X[grow * (long long)K + colbase + oct * 128 +
+                                   s * 16 + h * 8];
+            }
+        }
+    }
+}
+
 // Small-K sibling of v3 (K <= 512, multiple of 128): a whole 32-row x K
 // tile is ONE <=32-KB buffer, so the pipeline is a plain 3-buffer rotation
 // with prefetch depth TWO tiles (the K=1024 kernel must split tiles into
@@ -1662,7 +1701,17 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3s(
     }
 }
 
-template <int PROF>
+// v4 (TRB=1) image: X stored as row-major [4][16] subtiles so phase-B
+// fragments come from gfx950's hardware transpose-read (probe-verified
+// semantics, scripts/tr_probe.hip + gpurun_out/tr_probe.txt: each 16-lane
+// group hands lane (l&15) COLUMN l&15 of the 64-elem tile its addresses
+// cover, elems {base + (l&15) + 16j}).  Element offset of (row, col)
+// within a 32-row x 512-col half:
+__device__ __forceinline__ int v4_img(int row, int col) {
+    return (((row >> 2) * 32 + (col >> 4)) << 6) + ((row & 3) << 4) + (col & 15);
+}
+
+template <int PROF, int TRB>
 __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
     const unsigned short* __restrict__ X,   // [N][1024] bf16
     const unsigned short* __restrict__ y,   // [N] bf16
@@ -1725,8 +1774,13 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
     if (t_begin < t_end) {
         // prologue: stage tile t_begin fully.  h0 buffers alternate 0/2,
         // h1 lives in buffer 1.
-        v3_stage_half(X, n_rows, K, t_begin * V3_ROWS, 0, x_base, smem, wid, lane, nt_on);
-        v3_stage_half(X, n_rows, K, t_begin * V3_ROWS, V3_HALF, x_base + HBUF, smem, wid, lane, nt_on);
+        if constexpr (TRB) {
+            v4_stage_half(X, n_rows, K, t_begin * V3_ROWS, 0, x_base, smem, wid, lane, nt_on);
+            v4_stage_half(X, n_rows, K, t_begin * V3_ROWS, V3_HALF, x_base + HBUF, smem, wid, lane, nt_on);
+        } else {
+            v3_stage_half(X, n_rows, K, t_begin * V3_ROWS, 0, x_base, smem, wid, lane, nt_on);
+            v3_stage_half(X, n_rows, K, t_begin * V3_ROWS, V3_HALF, x_base + HBUF, smem, wid, lane, nt_on);
+        }
 
         int h0sel = 0;  // buffer index (0 or 2) holding the CURRENT tile's h0
 #pragma unroll 1
@@ -1754,8 +1808,12 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
             const unsigned yb0 = v3_load_y_asm(y + (yr0 > ymax ? ymax : yr0));
             const unsigned yb1 = v3_load_y_asm(y + (yr1 > ymax ? ymax : yr1));
             // [2] prefetch next tile's h0 as deep as possible
-            if (more)
-                v3_stage_half(X, n_rows, K, row0 + V3_ROWS, 0, h0n, smem, wid, lane, nt_on);
+            if (more) {
+                if constexpr (TRB)
+                    v4_stage_half(X, n_rows, K, row0 + V3_ROWS, 0, h0n, smem, wid, lane, nt_on);
+                else
+                    v3_stage_half(X, n_rows, K, row0 + V3_ROWS, 0, h0n, smem, wid, lane, nt_on);
+            }
 
             // ---- phase A on h0: z += X[:, w*128 .. +128) . theta ----
             unsigned long long pa = 0;
@@ -1768,8 +1826,13 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
                 const int g = (kc >> 3) + (lane >> 4);  // 16B group in half
                 const int arow0 = lane & 15;            // rows 0..15
                 const int arow1 = 16 + (lane & 15);     // rows 16..31
-                a0.q = *(U4*)&h0[arow0 * V3_HALF + ((g ^ v3_key(arow0)) * 8)];
-                a1.q = *(U4*)&h0[arow1 * V3_HALF + ((g ^ v3_key(arow1)) * 8)];
+                if constexpr (TRB) {
+                    a0.q = *(U4*)&h0[v4_img(arow0, g * 8)];
+                    a1.q = *(U4*)&h0[v4_img(arow1, g * 8)];
+                } else {
+                    a0.q = *(U4*)&h0[arow0 * V3_HALF + ((g ^ v3_key(arow0)) * 8)];
+                    a1.q = *(U4*)&h0[arow1 * V3_HALF + ((g ^ v3_key(arow1)) * 8)];
+                }
                 const int bk = kc + (lane >> 4) * 8;    // theta col (h0)
                 b0f.q = *(U4*)&th_lds[(lane & 15) * th_stride + bk];
                 z0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0.v, b0f.v, z0, 0, 0, 0);
@@ -1796,8 +1859,13 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
                 const int g = (kc >> 3) + (lane >> 4);
                 const int arow0 = lane & 15;
                 const int arow1 = 16 + (lane & 15);
-                a0.q = *(U4*)&h1[arow0 * V3_HALF + ((g ^ v3_key(arow0)) * 8)];
-                a1.q = *(U4*)&h1[arow1 * V3_HALF + ((g ^ v3_key(arow1)) * 8)];
+                if constexpr (TRB) {
+                    a0.q = *(U4*)&h1[v4_img(arow0, g * 8)];
+                    a1.q = *(U4*)&h1[v4_img(arow1, g * 8)];
+                } else {
+                    a0.q = *(U4*)&h1[arow0 * V3_HALF + ((g ^ v3_key(arow0)) * 8)];
+                    a1.q = *(U4*)&h1[arow1 * V3_HALF + ((g ^ v3_key(arow1)) * 8)];
+                }
                 const int bk = V3_HALF + kc + (lane >> 4) * 8;  // theta col (h1)
                 bf.q = *(U4*)&th_lds[(lane & 15) * th_stride + bk];
                 z0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0.v, bf.v, z0, 0, 0, 0);
@@ -1858,11 +1926,28 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
                 const int kcol = kc + (lane & 15);
                 f32x4_t acc = g_acc[8 + t2];
                 frag_u a, b;
+                if constexpr (TRB) {
+                    // two hardware transpose reads: lane (l&15) gets kcol
+                    // column kc+(l&15) for rows (l>>4)*8..+3 and +4..+7
+                    const unsigned base0 = (unsigned)((const char*)&h1[
+                        v4_img((lane >> 4) * 8, kc)] - smem) + (lane & 15) * 8u;
+                    const unsigned base1 = (unsigned)((const char*)&h1[
+                        v4_img((lane >> 4) * 8 + 4, kc)] - smem) + (lane & 15) * 8u;
+                    unsigned long long v0, v1;
+                    asm volatile(
+                        "ds_read_b64_tr_b16 %0, %2\n\t"
+                        "ds_read_b64_tr_b16 %1, %3\n\t"
+                        "s_waitcnt lgkmcnt(0)"
+                        : "=v"(v0), "=v"(v1) : "v"(base0), "v"(base1) : "memory");
+                    ((unsigned long long*)a.u)[0] = v0;
+                    ((unsigned long long*)a.u)[1] = v1;
+                } else {
 #pragma unroll
-                for (int j = 0; j < 8; ++j) {
-                    const int row = (lane >> 4) * 8 + j;
-                    a.u[j] = h1[row * V3_HALF +
-                                (((kcol >> 3) ^ v3_key(row)) * 8) + (kcol & 7)];
+                    for (int j = 0; j < 8; ++j) {
+                        const int row = (lane >> 4) * 8 + j;
+                        a.u[j] = h1[row * V3_HALF +
+                                    (((kcol >> 3) ^ v3_key(row)) * 8) + (kcol & 7)];
+                    }
                 }
                 b.q = *(U4*)&rt_lds[(lane & 15) * rt_stride + (lane >> 4) * 8];
                 g_acc[8 + t2] =
@@ -1876,8 +1961,12 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
             if (prof_on && wid == 0 && lane == 0)
                 p_bh += __builtin_amdgcn_s_memtime() - pt;
             // [6] refill h1 with the NEXT tile's second half
-            if (more)
-                v3_stage_half(X, n_rows, K, row0 + V3_ROWS, V3_HALF, h1, smem, wid, lane, nt_on);
+            if (more) {
+                if constexpr (TRB)
+                    v4_stage_half(X, n_rows, K, row0 + V3_ROWS, V3_HALF, h1, smem, wid, lane, nt_on);
+                else
+                    v3_stage_half(X, n_rows, K, row0 + V3_ROWS, V3_HALF, h1, smem, wid, lane, nt_on);
+            }
 
             if (prof_on && wid == 0 && lane == 0) pa = __builtin_amdgcn_s_memtime();
             // ---- phase B, h0 columns ----
@@ -1887,11 +1976,28 @@ __global__ __launch_bounds__(256) void k_logistic_glm_batched_v3(
                 const int kcol = kc + (lane & 15);
                 f32x4_t acc = g_acc[t2];
                 frag_u a, b;
+                if constexpr (TRB) {
+                    // two hardware transpose reads: lane (l&15) gets kcol
+                    // column kc+(l&15) for rows (l>>4)*8..+3 and +4..+7
+                    const unsigned base0 = (unsigned)((const char*)&h0[
+                        v4_img((lane >> 4) * 8, kc)] - smem) + (lane & 15) * 8u;
+                    const unsigned base1 = (unsigned)((const char*)&h0[
+                        v4_img((lane >> 4) * 8 + 4, kc)] - smem) + (lane & 15) * 8u;
+                    unsigned long long v0, v1;
+                    asm volatile(
+                        "ds_read_b64_tr_b16 %0, %2\n\t"
+                        "ds_read_b64_tr_b16 %1, %3\n\t"
+                        "s_waitcnt lgkmcnt(0)"
+                        : "=v"(v0), "=v"(v1) : "v"(base0), "v"(base1) : "memory");
+                    ((unsigned long long*)a.u)[0] = v0;
+                    ((unsigned long long*)a.u)[1] = v1;
+                } else {
 #pragma unroll
-                for (int j = 0; j < 8; ++j) {
-                    const int row = (lane >> 4) * 8 + j;
-                    a.u[j] = h0[row * V3_HALF +
-                                (((kcol >> 3) ^ v3_key(row)) * 8) + (kcol & 7)];
+                    for (int j = 0; j < 8; ++j) {
+                        const int row = (lane >> 4) * 8 + j;
+                        a.u[j] = h0[row * V3_HALF +
+                                    (((kcol >> 3) ^ v3_key(row)) * 8) + (kcol & 7)];
+                    }
                 }
                 b.q = *(U4*)&rt_lds[(lane & 15) * rt_stride + (lane >> 4) * 8];
                 g_acc[t2] =
@@ -2146,14 +2252,20 @@ extern "C" int fed_logistic_glm_batched(
         const char* pf = getenv("FED_V3_PROF");
         if (pf && atoi(pf) != 0) nt_on |= 2;
         if (K == 1024) {
-            if (nt_on & 2)
-                hipLaunchKernelGGL(k_logistic_glm_batched_v3<1>, dim3(g3), dim3(block),
-                                   lds3, stream, (const unsigned short*)X,
+            const char* v4e = getenv("FED_BATCHED_V4");
+            if (v4e && atoi(v4e) != 0)   // tr_b16 phase-B image (A/B)
+                hipLaunchKernelGGL((k_logistic_glm_batched_v3<0, 1>), dim3(g3),
+                                   dim3(block), lds3, stream, (const unsigned short*)X,
+                                   (const unsigned short*)y, n_rows,
+                                   (const unsigned short*)theta_t_bf16, workspace, nt_on);
+            else if (nt_on & 2)
+                hipLaunchKernelGGL((k_logistic_glm_batched_v3<1, 0>), dim3(g3),
+                                   dim3(block), lds3, stream, (const unsigned short*)X,
                                    (const unsigned short*)y, n_rows,
                                    (const unsigned short*)theta_t_bf16, workspace, nt_on);
             else
-                hipLaunchKernelGGL(k_logistic_glm_batched_v3<0>, dim3(g3), dim3(block),
-                                   lds3, stream, (const unsigned short*)X,
+                hipLaunchKernelGGL((k_logistic_glm_batched_v3<0, 0>), dim3(g3),
+                                   dim3(block), lds3, stream, (const unsigned short*)X,
                                    (const unsigned short*)y, n_rows,
                                    (const unsigned short*)theta_t_bf16, workspace, nt_on);
         }
