@@ -2252,8 +2252,11 @@ extern "C" int fed_logistic_glm_batched(
         const char* pf = getenv("FED_V3_PROF");
         if (pf && atoi(pf) != 0) nt_on |= 2;
         if (K == 1024) {
+            // tr_b16 phase-B image is the default: same-box A/B measured
+            // 0.888-0.891 vs 0.952-0.963 ms (3 reps, profiles raw_r2/v4_*);
+            // FED_BATCHED_V4=0 restores the scalar-gather image
             const char* v4e = getenv("FED_BATCHED_V4");
-            if (v4e && atoi(v4e) != 0)   // tr_b16 phase-B image (A/B)
+            if (!v4e || atoi(v4e) != 0)
                 hipLaunchKernelGGL((k_logistic_glm_batched_v3<0, 1>), dim3(g3),
                                    dim3(block), lds3, stream, (const unsigned short*)X,
                                    (const unsigned short*)y, n_rows,
