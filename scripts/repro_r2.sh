@@ -5,13 +5,14 @@ set -x
 cd "${GRAFT_REPO_ROOT:-/root/repo}"
 python -m pytensor_federated_amd.ops.build 2>&1 | tail -1
 python -m pytest tests -m gpu -q 2>&1 | grep -E "passed|failed" | tail -1
-# batched logistic (v3 default at K=1024: 0.878 ms @2e6x1024; 5.17 ms @config-4)
+# batched logistic (v4 tr_b16 default at K=1024: ~0.89 ms @2e6x1024; ~5.3 ms @config-4)
 python benchmarks/bench_batched_chains.py --rows 2000000 --steps 50 | tail -1
 python benchmarks/bench_batched_chains.py --rows 12500000 --steps 40 | tail -1
 python benchmarks/bench_batched_chains.py --rows 2000000 --features 512 --steps 50 | tail -1
-# ladder A/Bs
+# ladder A/Bs (v1 / v2 / v3-scalar-image / v4-default)
 FED_BATCHED_V1=1 python benchmarks/bench_batched_chains.py --rows 2000000 --steps 50 | tail -1
 FED_BATCHED_V3=0 python benchmarks/bench_batched_chains.py --rows 2000000 --steps 50 | tail -1
+FED_BATCHED_V4=0 python benchmarks/bench_batched_chains.py --rows 2000000 --steps 50 | tail -1
 # windowed (default) vs simple NUTS adaptation
 python benchmarks/bench_nuts_batched.py --chains 16 --draws 300 --tune 400 --mass dense | tail -1
 python benchmarks/bench_nuts_batched.py --chains 16 --draws 300 --tune 400 --mass dense --adaptation simple | tail -1
